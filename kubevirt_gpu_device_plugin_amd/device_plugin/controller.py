@@ -1,0 +1,120 @@
+"""Controller: discovery → plugin creation → lifetime.
+
+AMD equivalent of the reference controller
+(reference: InitiateDevicePlugin/createDevicePlugins,
+device_plugin.go:89-176): one passthrough plugin per GPU PCI device id,
+one VF plugin per VF PCI device id, each started independently (a
+per-type start failure is logged and the others continue,
+device_plugin.go:131-136), then block until stopped.
+"""
+
+import logging
+import threading
+
+from .. import dpapi
+from ..topology import build_island_lookup
+from . import consts, discovery, pciids
+from .plugin import GenericDevicePlugin
+from .plugin_base import PluginConfig
+from .vf_plugin import VfDevicePlugin
+
+log = logging.getLogger(__name__)
+
+
+def build_kubelet_devices(devs):
+    """pluginapi.Device list with NUMA topology
+    (reference: device_plugin.go:109-123)."""
+    out = []
+    for d in devs:
+        out.append(dpapi.Device(
+            ID=d.addr, health=dpapi.HEALTHY,
+            topology=dpapi.TopologyInfo(
+                nodes=[dpapi.NUMANode(ID=d.numa_node)])))
+    return out
+
+
+def resolve_name(device_id, pci_ids_path=None):
+    """Marketing name from pci.ids, raw hex id as fallback
+    (reference: device_plugin.go:124-128)."""
+    name = pciids.get_device_name(device_id, pci_ids_path=pci_ids_path)
+    if not name:
+        log.error("could not find device name for device id %s; using "
+                  "raw id", device_id)
+        return device_id
+    return name
+
+
+class Controller:
+    """Owns discovery output and the per-resource plugin servers."""
+
+    def __init__(self, config=None, pci_ids_path=None,
+                 kfd_nodes_dir=consts.KFD_TOPOLOGY_PATH,
+                 vf_event_watcher_factory=None):
+        self.config = config or PluginConfig()
+        self.pci_ids_path = pci_ids_path
+        self.kfd_nodes_dir = kfd_nodes_dir
+        self.vf_event_watcher_factory = vf_event_watcher_factory
+        self.registry = None
+        self.plugins = []
+
+    def create_plugins(self):
+        """Run discovery and instantiate (but not start) all plugin
+        servers.  Returns the plugin list."""
+        self.registry = discovery.discover(base_path=self.config.pci_base)
+        island_of = build_island_lookup(
+            self.registry, nodes_dir=self.kfd_nodes_dir)
+        log.info("iommu map: %s",
+                 {g: [d.addr for d in v]
+                  for g, v in self.registry.iommu_map.items()})
+        log.info("device map: %s",
+                 {k: [d.addr for d in v]
+                  for k, v in self.registry.device_map.items()})
+        log.info("vf map: %s",
+                 {k: [d.addr for d in v]
+                  for k, v in self.registry.vf_map.items()})
+        log.info("pf→vf map: %s", self.registry.pf_vf_map)
+
+        self.plugins = []
+        for device_id, devs in sorted(self.registry.device_map.items()):
+            name = resolve_name(device_id, self.pci_ids_path)
+            self.plugins.append(GenericDevicePlugin(
+                name, build_kubelet_devices(devs), self.registry,
+                config=self.config, island_of=island_of))
+        for device_id, devs in sorted(self.registry.vf_map.items()):
+            name = resolve_name(device_id, self.pci_ids_path)
+            self.plugins.append(VfDevicePlugin(
+                name, build_kubelet_devices(devs), self.registry,
+                config=self.config, island_of=island_of,
+                event_watcher_factory=self.vf_event_watcher_factory))
+        return self.plugins
+
+    def start(self, stop_event):
+        started = []
+        for p in self.plugins:
+            try:
+                p.start(stop_event)
+                started.append(p)
+            except Exception as e:
+                log.error("error starting %s device plugin: %s",
+                          p.device_name, e)
+        self.plugins = started
+        return started
+
+    def stop(self):
+        for p in self.plugins:
+            try:
+                p.stop()
+            except Exception:
+                log.exception("error stopping %s", p.device_name)
+
+
+def initiate_device_plugin(stop_event=None, **kwargs):
+    """Blocking entry point (reference: InitiateDevicePlugin,
+    device_plugin.go:89-96)."""
+    stop_event = stop_event or threading.Event()
+    controller = Controller(**kwargs)
+    controller.create_plugins()
+    controller.start(stop_event)
+    stop_event.wait()
+    log.info("shutting down device plugin controller")
+    controller.stop()

@@ -1,0 +1,257 @@
+"""Common DevicePlugin server lifecycle.
+
+Python/grpcio equivalent of the reference's per-resource gRPC server:
+socket creation, readiness self-dial, kubelet registration, ListAndWatch
+health streaming, socket-removal restart
+(reference: generic_device_plugin.go:217-351,619-697).
+
+Differences from the reference, by design:
+  * health transitions go through a lock+condition-versioned device list
+    instead of unbuffered channels, so ListAndWatch reconnects (kubelet
+    restarts) always see current state and health producers never block;
+  * one inotify watch on the /dev/vfio directory covers every group node
+    (the reference adds one fsnotify watch per device node and cannot see
+    re-creation of a removed node, generic_device_plugin.go:647-678);
+  * after a kubelet restart the restarted server keeps the controller's
+    stop signal (the reference's restarted server detaches from the
+    global stop channel, generic_device_plugin.go:283-285).
+"""
+
+import logging
+import os
+import threading
+from concurrent import futures
+from dataclasses import dataclass
+
+import grpc
+
+from .. import dpapi
+from . import consts, inotify
+
+log = logging.getLogger(__name__)
+
+
+@dataclass
+class PluginConfig:
+    """Host paths, injectable for tests (takes the role of the
+    reference's rebindable package globals, device_plugin.go:70-79)."""
+    device_plugin_dir: str = dpapi.DEVICE_PLUGIN_PATH
+    kubelet_socket: str = dpapi.KUBELET_SOCKET
+    vfio_dir: str = consts.VFIO_DEVICE_PATH
+    iommu_dev: str = consts.IOMMU_DEVICE_PATH
+    pci_base: str = consts.PCI_DEVICES_PATH
+    namespace: str = consts.DEVICE_NAMESPACE
+    connect_timeout_s: float = consts.CONNECTION_TIMEOUT_S
+
+
+class DevicePluginBase(dpapi.DevicePluginServicer):
+    """One kubelet-facing gRPC server per resource type."""
+
+    def __init__(self, device_name, devices, config=None):
+        self.device_name = device_name
+        self.config = config or PluginConfig()
+        self.socket_path = os.path.join(
+            self.config.device_plugin_dir,
+            "kubevirt-%s.sock" % device_name)
+        self._devs = list(devices)   # dpapi.Device messages (mutable)
+        self._lock = threading.Condition()
+        self._version = 0
+        self._server = None
+        self._stop = None            # controller-owned threading.Event
+        self._term = threading.Event()
+        self._health_thread = None
+
+    # ---- lifecycle ------------------------------------------------------
+
+    @property
+    def resource_name(self):
+        return "%s/%s" % (self.config.namespace, self.device_name)
+
+    def start(self, stop_event):
+        """Create the socket, serve, self-dial, register with kubelet and
+        start health watching (reference: Start, generic_device_plugin.go:217-257)."""
+        if self._server is not None:
+            raise RuntimeError("gRPC server already started")
+        self._stop = stop_event
+        self._term = threading.Event()
+        self._cleanup_socket()
+
+        server = grpc.server(
+            futures.ThreadPoolExecutor(max_workers=8),
+            options=(("grpc.so_reuseport", 0),))
+        dpapi.add_device_plugin_servicer(self, server)
+        server.add_insecure_port("unix:" + self.socket_path)
+        server.start()
+        self._server = server
+
+        self._wait_for_ready()
+        try:
+            self.register()
+        except grpc.RpcError as e:
+            log.error("[%s] error registering with kubelet: %s",
+                      self.device_name, e)
+            self.stop()
+            raise
+
+        self._health_thread = threading.Thread(
+            target=self._health_loop_guard,
+            name="health-%s" % self.device_name, daemon=True)
+        self._health_thread.start()
+        log.info("%s device plugin server ready", self.device_name)
+
+    def stop(self):
+        if self._server is None:
+            return
+        self._term.set()
+        with self._lock:
+            self._lock.notify_all()
+        server, self._server = self._server, None
+        server.stop(grace=None)
+        self._cleanup_socket()
+
+    def restart(self):
+        """Full re-handshake after a kubelet restart
+        (reference: restart, generic_device_plugin.go:275-286)."""
+        log.info("restarting %s device plugin server", self.device_name)
+        stop_event = self._stop
+        self.stop()
+        self.start(stop_event)
+
+    def _cleanup_socket(self):
+        try:
+            os.remove(self.socket_path)
+        except FileNotFoundError:
+            pass
+
+    def _wait_for_ready(self):
+        ch = grpc.insecure_channel("unix:" + self.socket_path)
+        try:
+            grpc.channel_ready_future(ch).result(
+                timeout=self.config.connect_timeout_s)
+        finally:
+            ch.close()
+
+    def register(self):
+        """One-shot Registration RPC into kubelet
+        (reference: Register, generic_device_plugin.go:289-310)."""
+        ch = grpc.insecure_channel("unix:" + self.config.kubelet_socket)
+        try:
+            grpc.channel_ready_future(ch).result(
+                timeout=self.config.connect_timeout_s)
+            stub = dpapi.RegistrationStub(ch)
+            stub.Register(
+                dpapi.RegisterRequest(
+                    version=dpapi.VERSION,
+                    endpoint=os.path.basename(self.socket_path),
+                    resource_name=self.resource_name),
+                timeout=self.config.connect_timeout_s)
+        finally:
+            ch.close()
+        log.info("[%s] registered with kubelet for resource %s",
+                 self.device_name, self.resource_name)
+
+    # ---- device state ---------------------------------------------------
+
+    def devices_snapshot(self):
+        with self._lock:
+            return [dpapi.Device.FromString(d.SerializeToString())
+                    for d in self._devs]
+
+    def set_health(self, device_ids, health):
+        """Flip health for the given advertised IDs and wake ListAndWatch
+        streams (takes the role of the healthy/unhealthy channels,
+        generic_device_plugin.go:326-343)."""
+        if isinstance(device_ids, str):
+            device_ids = [device_ids]
+        wanted = set(device_ids)
+        with self._lock:
+            changed = False
+            for d in self._devs:
+                if d.ID in wanted and d.health != health:
+                    d.health = health
+                    changed = True
+            if changed:
+                self._version += 1
+                self._lock.notify_all()
+
+    # ---- DevicePlugin service ------------------------------------------
+
+    def GetDevicePluginOptions(self, request, context):  # noqa: N802
+        return dpapi.DevicePluginOptions(
+            pre_start_required=False,
+            get_preferred_allocation_available=True)
+
+    def PreStartContainer(self, request, context):  # noqa: N802
+        return dpapi.PreStartContainerResponse()
+
+    def ListAndWatch(self, request, context):  # noqa: N802
+        """Initial full list, then a fresh full list on every health
+        transition (reference: generic_device_plugin.go:313-350)."""
+        log.info("[%s] ListAndWatch: sending %d devices",
+                 self.device_name, len(self._devs))
+        with self._lock:
+            last = self._version
+            yield dpapi.ListAndWatchResponse(devices=self._devs)
+        while context.is_active():
+            with self._lock:
+                if self._version == last and not self._should_exit():
+                    self._lock.wait(timeout=0.5)
+                if self._should_exit():
+                    return
+                if self._version != last:
+                    last = self._version
+                    yield dpapi.ListAndWatchResponse(devices=self._devs)
+
+    def _should_exit(self):
+        return self._term.is_set() or (self._stop is not None
+                                       and self._stop.is_set())
+
+    # ---- health ---------------------------------------------------------
+
+    def _health_loop_guard(self):
+        try:
+            self._health_loop()
+        except Exception:
+            log.exception("[%s] health loop failed", self.device_name)
+
+    def _group_to_ids(self):
+        """Map watched /dev/vfio node names to advertised device IDs.
+        Overridden by subclasses that know the registry."""
+        return {}
+
+    def _health_loop(self):
+        """inotify loop: vfio node create/remove → health flips; removal
+        of our own socket → kubelet restarted → full server restart
+        (reference: healthCheck, generic_device_plugin.go:619-697)."""
+        group_to_ids = self._group_to_ids()
+        sock_base = os.path.basename(self.socket_path)
+        with inotify.Watcher() as w:
+            w.add_watch(self.config.device_plugin_dir)
+            try:
+                w.add_watch(self.config.vfio_dir)
+            except OSError as e:
+                log.warning("[%s] cannot watch %s: %s", self.device_name,
+                            self.config.vfio_dir, e)
+            while not self._should_exit():
+                for ev in w.read_events(timeout_s=0.2):
+                    path = w.path_of(ev.wd)
+                    if (path == self.config.device_plugin_dir
+                            and ev.name == sock_base
+                            and ev.mask & (inotify.IN_DELETE
+                                           | inotify.IN_MOVED_FROM)):
+                        log.info("[%s] socket removed; kubelet likely "
+                                 "restarted", self.device_name)
+                        self.restart()
+                        return
+                    if path == self.config.vfio_dir \
+                            and ev.name in group_to_ids:
+                        ids = group_to_ids[ev.name]
+                        if ev.mask & (inotify.IN_CREATE
+                                      | inotify.IN_MOVED_TO):
+                            self.set_health(ids, dpapi.HEALTHY)
+                        elif ev.mask & (inotify.IN_DELETE
+                                        | inotify.IN_MOVED_FROM):
+                            log.info("[%s] vfio node %s removed; marking "
+                                     "%s unhealthy", self.device_name,
+                                     ev.name, ids)
+                            self.set_health(ids, dpapi.UNHEALTHY)
