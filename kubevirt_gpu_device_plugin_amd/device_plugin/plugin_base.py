@@ -59,6 +59,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         self._server = None
         self._stop = None            # controller-owned threading.Event
         self._term = threading.Event()
+        self._watch_armed = threading.Event()
         self._health_thread = None
 
     # ---- lifecycle ------------------------------------------------------
@@ -93,10 +94,14 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             self.stop()
             raise
 
+        self._watch_armed = threading.Event()
         self._health_thread = threading.Thread(
             target=self._health_loop_guard,
             name="health-%s" % self.device_name, daemon=True)
         self._health_thread.start()
+        # Don't report ready until the watcher is armed, so a socket
+        # removal immediately after start is never missed.
+        self._watch_armed.wait(timeout=2.0)
         log.info("%s device plugin server ready", self.device_name)
 
     def stop(self):
@@ -232,8 +237,11 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             except OSError as e:
                 log.warning("[%s] cannot watch %s: %s", self.device_name,
                             self.config.vfio_dir, e)
+            self._watch_armed.set()
             while not self._should_exit():
                 for ev in w.read_events(timeout_s=0.2):
+                    if self._should_exit():
+                        return
                     path = w.path_of(ev.wd)
                     if (path == self.config.device_plugin_dir
                             and ev.name == sock_base
