@@ -1,0 +1,177 @@
+// _healthprobe — gfx950 GPU self-test kernel.
+//
+// The plugin's host-side GPU sanity check for nodes where a PF is
+// host-driver-resident (SR-IOV): before VFs of a GPU are advertised
+// Healthy, a quick HBM write/readback at bandwidth plus an MFMA-unit
+// exercise confirms the silicon answers.  The reference has no GPU
+// compute at all (SURVEY.md §2 — its only native code is the NVML
+// binding); this is the MI355X-native extra that makes health checking
+// mean more than "the /dev node exists".
+//
+// CDNA4 notes (per /opt/skills/guides/MI355X_MICROARCH.md): 64-wide
+// wavefronts, 256 threads/block, grid-stride loops sized ≫256
+// workgroups to cover all 8 XCDs; dwordx4 vector traffic for HBM3E.
+
+#include <pybind11/pybind11.h>
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                   \
+  do {                                                                    \
+    hipError_t _e = (expr);                                               \
+    if (_e != hipSuccess)                                                 \
+      throw std::runtime_error(std::string(#expr) + ": " +                \
+                               hipGetErrorString(_e));                    \
+  } while (0)
+
+namespace {
+
+// Pattern fill: each lane writes a value derived from its index so the
+// verify pass can detect addressing faults, not just stuck bits.
+__global__ void fill_kernel(uint4 *__restrict__ out, size_t n_vec,
+                            uint32_t seed) {
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = gridDim.x * blockDim.x;
+  for (; i < n_vec; i += stride) {
+    uint32_t base = seed ^ (uint32_t)(i * 2654435761u);
+    out[i] = make_uint4(base, base + 1, base + 2, base + 3);
+  }
+}
+
+// Verify + bandwidth read pass: XOR-reduce everything; mismatches flip
+// the error counter.
+__global__ void verify_kernel(const uint4 *__restrict__ in, size_t n_vec,
+                              uint32_t seed,
+                              unsigned long long *__restrict__ errors) {
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = gridDim.x * blockDim.x;
+  unsigned long long local = 0;
+  for (; i < n_vec; i += stride) {
+    uint32_t base = seed ^ (uint32_t)(i * 2654435761u);
+    uint4 v = in[i];
+    local += (v.x != base) + (v.y != base + 1) + (v.z != base + 2) +
+             (v.w != base + 3);
+  }
+  if (local) atomicAdd(errors, local);
+}
+
+// Minimal matrix-core exercise: one MFMA per wavefront, result checked
+// on host.  Confirms the XCD compute path beyond plain VALU/HBM.
+__global__ void mfma_kernel(float *__restrict__ out) {
+#if defined(__gfx950__) || defined(__gfx942__) || defined(__gfx90a__)
+  using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  // C += A*B with A=B=1 on a 16x16x4 tile: every output element = 4.
+  acc = __builtin_amdgcn_mfma_f32_16x16x4f32(1.0f, 1.0f, acc, 0, 0, 0);
+  int lane = threadIdx.x & 63;
+  if (blockIdx.x == 0 && threadIdx.x < 64) {
+    out[lane * 4 + 0] = acc[0];
+    out[lane * 4 + 1] = acc[1];
+    out[lane * 4 + 2] = acc[2];
+    out[lane * 4 + 3] = acc[3];
+  }
+#else
+  if (blockIdx.x == 0 && threadIdx.x == 0) out[0] = -1.0f;
+#endif
+}
+
+py::dict probe(int device, size_t mib) {
+  py::dict result;
+  int ndev = 0;
+  HIP_CHECK(hipGetDeviceCount(&ndev));
+  result["device_count"] = ndev;
+  if (device >= ndev)
+    throw std::runtime_error("device index out of range");
+  HIP_CHECK(hipSetDevice(device));
+
+  hipDeviceProp_t prop;
+  HIP_CHECK(hipGetDeviceProperties(&prop, device));
+  result["name"] = std::string(prop.name);
+  result["gcn_arch"] = std::string(prop.gcnArchName);
+  result["vram_gib"] =
+      static_cast<double>(prop.totalGlobalMem) / (1 << 30);
+  result["multi_processor_count"] = prop.multiProcessorCount;
+
+  size_t bytes = mib << 20;
+  size_t n_vec = bytes / sizeof(uint4);
+  uint4 *buf = nullptr;
+  HIP_CHECK(hipMalloc(&buf, bytes));
+  unsigned long long *errors = nullptr;
+  HIP_CHECK(hipMalloc(&errors, sizeof(unsigned long long)));
+  HIP_CHECK(hipMemset(errors, 0, sizeof(unsigned long long)));
+
+  // ≫256 workgroups so all 8 XCDs are covered.
+  int blocks = prop.multiProcessorCount * 8;
+  dim3 grid(blocks), block(256);
+  uint32_t seed = 0xA31DBEEF;
+
+  hipEvent_t t0, t1, t2;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventCreate(&t2));
+
+  // warmup
+  hipLaunchKernelGGL(fill_kernel, grid, block, 0, 0, buf, n_vec, seed);
+  HIP_CHECK(hipDeviceSynchronize());
+
+  HIP_CHECK(hipEventRecord(t0));
+  hipLaunchKernelGGL(fill_kernel, grid, block, 0, 0, buf, n_vec, seed);
+  HIP_CHECK(hipEventRecord(t1));
+  hipLaunchKernelGGL(verify_kernel, grid, block, 0, 0, buf, n_vec, seed,
+                     errors);
+  HIP_CHECK(hipEventRecord(t2));
+  HIP_CHECK(hipEventSynchronize(t2));
+
+  float fill_ms = 0, verify_ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&fill_ms, t0, t1));
+  HIP_CHECK(hipEventElapsedTime(&verify_ms, t1, t2));
+  unsigned long long h_errors = 0;
+  HIP_CHECK(hipMemcpy(&h_errors, errors, sizeof h_errors,
+                      hipMemcpyDeviceToHost));
+
+  float *mfma_out = nullptr;
+  HIP_CHECK(hipMalloc(&mfma_out, 256 * sizeof(float)));
+  HIP_CHECK(hipMemset(mfma_out, 0, 256 * sizeof(float)));
+  hipLaunchKernelGGL(mfma_kernel, dim3(1), dim3(64), 0, 0, mfma_out);
+  HIP_CHECK(hipDeviceSynchronize());
+  float h_mfma[256];
+  HIP_CHECK(hipMemcpy(h_mfma, mfma_out, sizeof h_mfma,
+                      hipMemcpyDeviceToHost));
+  bool mfma_ok = true;
+  for (int i = 0; i < 256; ++i)
+    if (h_mfma[i] != 4.0f) mfma_ok = false;
+
+  HIP_CHECK(hipFree(mfma_out));
+  HIP_CHECK(hipFree(errors));
+  HIP_CHECK(hipFree(buf));
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  hipEventDestroy(t2);
+
+  double gib = static_cast<double>(bytes) / (1 << 30);
+  result["bytes"] = bytes;
+  result["write_gbps"] = gib / (fill_ms / 1e3);
+  result["read_gbps"] = gib / (verify_ms / 1e3);
+  result["pattern_errors"] = h_errors;
+  result["mfma_ok"] = mfma_ok;
+  result["ok"] = (h_errors == 0) && mfma_ok;
+  return result;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_healthprobe, m) {
+  m.doc() = "gfx950 GPU self-test: HBM pattern + bandwidth + MFMA";
+  m.def("probe", &probe, py::arg("device") = 0, py::arg("mib") = 1024);
+  m.def("device_count", [] {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+  });
+}

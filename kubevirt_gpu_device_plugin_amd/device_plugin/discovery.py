@@ -67,10 +67,48 @@ class DeviceRegistry:
             yield from devs
 
 
+def _native_scan(base_path):
+    """One-pass scan via the C++ _sysfs extension; None when the
+    extension is unavailable (pure-Python walk takes over)."""
+    try:
+        from kubevirt_gpu_device_plugin_amd import _sysfs
+    except ImportError:
+        return None
+    try:
+        return _sysfs.scan_pci(base_path, consts.AMD_VENDOR_ID)
+    except RuntimeError as e:
+        log.error("native scan of %s failed: %s", base_path, e)
+        return None
+
+
 def discover(base_path=consts.PCI_DEVICES_PATH,
-             supported_drivers=consts.SUPPORTED_VFIO_DRIVERS):
-    """Walk the PCI bus once and build the :class:`DeviceRegistry`."""
+             supported_drivers=consts.SUPPORTED_VFIO_DRIVERS,
+             use_native=None):
+    """Walk the PCI bus once and build the :class:`DeviceRegistry`.
+
+    ``use_native``: True forces the C++ scanner, False the Python walk,
+    None picks native when built.  Both produce identical registries
+    (pinned by tests/test_discovery.py::test_native_python_parity).
+    """
     reg = DeviceRegistry()
+    records = _native_scan(base_path) if use_native in (None, True) \
+        else None
+    if use_native is True and records is None:
+        raise RuntimeError("native _sysfs extension required but "
+                           "unavailable")
+    if records is not None:
+        for r in records:
+            if r["driver"] not in supported_drivers:
+                log.info("skipping %s: driver %s is not a supported "
+                         "vfio driver", r["addr"], r["driver"])
+                continue
+            dev = AmdGpuDevice(addr=r["addr"], numa_node=r["numa_node"],
+                               device_id=r["device"],
+                               iommu_group=r["iommu_group"],
+                               parent_pf=r["physfn"])
+            _register(reg, dev)
+        return reg
+
     try:
         entries = sorted(os.listdir(base_path))
     except OSError as e:
@@ -116,16 +154,20 @@ def discover(base_path=consts.PCI_DEVICES_PATH,
         dev = AmdGpuDevice(addr=addr, numa_node=numa_node,
                            device_id=device_id, iommu_group=iommu_group,
                            parent_pf=parent_pf)
-        reg.iommu_map.setdefault(iommu_group, []).append(dev)
-        reg.bdf_to_iommu[addr] = iommu_group
-        if dev.is_vf:
-            reg.vf_map.setdefault(device_id, []).append(dev)
-            reg.pf_vf_map.setdefault(parent_pf, []).append(addr)
-            log.info("discovered AMD VF %s (type %s, PF %s, iommu %s, "
-                     "numa %d)", addr, device_id, parent_pf, iommu_group,
-                     numa_node)
-        else:
-            reg.device_map.setdefault(device_id, []).append(dev)
-            log.info("discovered AMD GPU %s (type %s, iommu %s, numa %d)",
-                     addr, device_id, iommu_group, numa_node)
+        _register(reg, dev)
     return reg
+
+
+def _register(reg, dev):
+    reg.iommu_map.setdefault(dev.iommu_group, []).append(dev)
+    reg.bdf_to_iommu[dev.addr] = dev.iommu_group
+    if dev.is_vf:
+        reg.vf_map.setdefault(dev.device_id, []).append(dev)
+        reg.pf_vf_map.setdefault(dev.parent_pf, []).append(dev.addr)
+        log.info("discovered AMD VF %s (type %s, PF %s, iommu %s, "
+                 "numa %d)", dev.addr, dev.device_id, dev.parent_pf,
+                 dev.iommu_group, dev.numa_node)
+    else:
+        reg.device_map.setdefault(dev.device_id, []).append(dev)
+        log.info("discovered AMD GPU %s (type %s, iommu %s, numa %d)",
+                 dev.addr, dev.device_id, dev.iommu_group, dev.numa_node)

@@ -23,6 +23,11 @@ from . import consts, sysfs
 from .allocation import preferred_allocation
 from .plugin_base import DevicePluginBase
 
+try:  # hot-path revalidation in C++ (csrc/sysfs_scan.cpp)
+    from kubevirt_gpu_device_plugin_amd import _sysfs
+except ImportError:
+    _sysfs = None
+
 log = logging.getLogger(__name__)
 
 
@@ -102,43 +107,60 @@ class GenericDevicePlugin(DevicePluginBase):
                         grpc.StatusCode.INVALID_ARGUMENT,
                         "invalid allocation request: unknown device: %s"
                         % bdf)
-                requested_found = False
-                for dev in members:
-                    try:
-                        live_group = sysfs.read_link_basename(
-                            base, dev.addr, "iommu_group")
-                    except OSError:
-                        live_group = None
-                    if live_group != group:
-                        log.warning("IOMMU group changed for %s",
-                                    dev.addr)
+                requested_found = any(d.addr == bdf for d in members)
+                if _sysfs is not None:
+                    # One C call revalidates the whole group and (for
+                    # iommufd) collects the cdev names.
+                    failed, vfio_devs = _sysfs.revalidate(
+                        base, [(d.addr, group) for d in members],
+                        consts.AMD_VENDOR_ID, iommufd)
+                    if failed:
+                        log.warning("revalidation failed for %s", failed)
                         context.abort(
                             grpc.StatusCode.INVALID_ARGUMENT,
                             "invalid allocation request: unknown "
-                            "device: %s" % dev.addr)
-                    try:
-                        vendor = sysfs.read_id_from_file(
-                            base, dev.addr, "vendor")
-                    except OSError:
-                        vendor = None
-                    if vendor != consts.AMD_VENDOR_ID:
-                        log.warning("vendor changed for %s", dev.addr)
-                        context.abort(
-                            grpc.StatusCode.INVALID_ARGUMENT,
-                            "invalid allocation request: unknown "
-                            "device: %s" % dev.addr)
-                    if dev.addr == bdf:
-                        requested_found = True
+                            "device: %s" % failed)
                     if iommufd:
+                        for cdev in vfio_devs:
+                            add_spec(posixpath_join(
+                                self.config.vfio_dir, "devices", cdev))
+                else:
+                    for dev in members:
                         try:
-                            vfiodev = sysfs.read_vfio_dev(base, dev.addr)
+                            live_group = sysfs.read_link_basename(
+                                base, dev.addr, "iommu_group")
                         except OSError:
+                            live_group = None
+                        if live_group != group:
+                            log.warning("IOMMU group changed for %s",
+                                        dev.addr)
                             context.abort(
-                                grpc.StatusCode.INTERNAL,
-                                "could not determine iommufd device "
-                                "for device %s" % dev.addr)
-                        add_spec(posixpath_join(
-                            self.config.vfio_dir, "devices", vfiodev))
+                                grpc.StatusCode.INVALID_ARGUMENT,
+                                "invalid allocation request: unknown "
+                                "device: %s" % dev.addr)
+                        try:
+                            vendor = sysfs.read_id_from_file(
+                                base, dev.addr, "vendor")
+                        except OSError:
+                            vendor = None
+                        if vendor != consts.AMD_VENDOR_ID:
+                            log.warning("vendor changed for %s", dev.addr)
+                            context.abort(
+                                grpc.StatusCode.INVALID_ARGUMENT,
+                                "invalid allocation request: unknown "
+                                "device: %s" % dev.addr)
+                        if iommufd:
+                            try:
+                                vfiodev = sysfs.read_vfio_dev(
+                                    base, dev.addr)
+                            except OSError:
+                                context.abort(
+                                    grpc.StatusCode.INTERNAL,
+                                    "could not determine iommufd device "
+                                    "for device %s" % dev.addr)
+                            add_spec(posixpath_join(
+                                self.config.vfio_dir, "devices",
+                                vfiodev))
                 if not requested_found:
                     context.abort(
                         grpc.StatusCode.INVALID_ARGUMENT,

@@ -100,3 +100,28 @@ def test_mi355x_node_scale(synthetic_host):
     assert len(reg.vf_map["75b3"]) == 64
     assert len(reg.pf_vf_map) == 8
     assert all(len(v) == 8 for v in reg.pf_vf_map.values())
+
+
+def test_native_python_parity(synthetic_host):
+    """The C++ scanner and the Python walk must build identical
+    registries."""
+    import pytest
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", iommu_group="100", numa=1)
+    h.add_gpu("0000:11:00.0", iommu_group="100", numa=1)  # co-grouped
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0", iommu_group="120")
+    h.add_pci_device("0000:30:00.0", vendor="10de")
+    try:
+        reg_native = discovery.discover(base_path=h.pci, use_native=True)
+    except RuntimeError:
+        pytest.skip("native _sysfs extension not built")
+    reg_py = discovery.discover(base_path=h.pci, use_native=False)
+    assert reg_native.bdf_to_iommu == reg_py.bdf_to_iommu
+    assert {k: [d.addr for d in v] for k, v in reg_native.iommu_map.items()} \
+        == {k: [d.addr for d in v] for k, v in reg_py.iommu_map.items()}
+    assert {k: [(d.addr, d.numa_node) for d in v]
+            for k, v in reg_native.device_map.items()} \
+        == {k: [(d.addr, d.numa_node) for d in v]
+            for k, v in reg_py.device_map.items()}
+    assert reg_native.pf_vf_map == reg_py.pf_vf_map

@@ -1,0 +1,140 @@
+"""Tests that need a real MI355X box (run via gpurun).
+
+These exercise the native paths end-to-end: the gfx950 HIP health-probe
+kernel, the AMD-SMI dlopen binding against the live driver, and KFD
+topology parsing on real sysfs.  They fail loudly when the native
+extensions are missing — a silent fallback would make GPU health
+checking meaningless.
+"""
+
+import os
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def _hip_device_count():
+    from kubevirt_gpu_device_plugin_amd import _healthprobe
+    return _healthprobe.device_count()
+
+
+@pytest.fixture(scope="module", autouse=True)
+def require_gpu():
+    # import error here = native code not built → loud failure, no skip
+    from kubevirt_gpu_device_plugin_amd import _healthprobe  # noqa: F401
+    if _hip_device_count() < 1:
+        pytest.fail("gpu-marked test run without a visible HIP device")
+
+
+def test_native_extensions_are_in_tree():
+    import kubevirt_gpu_device_plugin_amd as pkg
+    from kubevirt_gpu_device_plugin_amd import _amdsmi, _healthprobe
+    pkg_dir = os.path.dirname(os.path.abspath(pkg.__file__))
+    assert os.path.dirname(os.path.abspath(
+        _amdsmi.__file__)) == pkg_dir
+    assert os.path.dirname(os.path.abspath(
+        _healthprobe.__file__)) == pkg_dir
+
+
+def test_health_probe_kernel():
+    """HBM pattern + bandwidth + MFMA on the real gfx950."""
+    from kubevirt_gpu_device_plugin_amd import _healthprobe
+    r = _healthprobe.probe(0, 1024)
+    assert r["pattern_errors"] == 0
+    assert r["mfma_ok"], "MFMA 16x16x4 f32 result mismatch"
+    assert r["ok"]
+    assert "gfx950" in r["gcn_arch"], r["gcn_arch"]
+    # MI355X: 288 GB HBM3E, ≈8 TB/s peak; a grid-stride uint4 stream
+    # should comfortably exceed 2 TB/s on both passes
+    assert r["vram_gib"] > 200, r["vram_gib"]
+    assert r["write_gbps"] > 2000, r
+    assert r["read_gbps"] > 2000, r
+
+
+def test_amdsmi_binding_enumerates():
+    from kubevirt_gpu_device_plugin_amd import _amdsmi
+    assert _amdsmi.available()
+    _amdsmi.init()
+    try:
+        devs = _amdsmi.get_devices()
+        assert len(devs) >= 1
+        d = devs[0]
+        assert len(d["bdf"].split(":")) == 3, d
+        assert d["uuid"], d
+        # MI355X is a 256-CU part
+        if "num_compute_units" in d:
+            assert d["num_compute_units"] >= 200, d
+    finally:
+        _amdsmi.shutdown()
+
+
+def test_amdsmi_xgmi_info():
+    from kubevirt_gpu_device_plugin_amd import _amdsmi
+    _amdsmi.init()
+    try:
+        info = _amdsmi.xgmi_info(0)
+        assert "hive_id" in info and "node_id" in info
+    except RuntimeError as e:
+        # single-GPU boxes may report no hive — that's valid data
+        pytest.skip("xgmi_info unsupported here: %s" % e)
+    finally:
+        _amdsmi.shutdown()
+
+
+def test_amdsmi_ecc_count():
+    from kubevirt_gpu_device_plugin_amd import _amdsmi
+    _amdsmi.init()
+    try:
+        ec = _amdsmi.ecc_count(0)
+        assert ec["uncorrectable"] == 0, "GPU reports RAS errors: %r" % ec
+    except RuntimeError as e:
+        pytest.skip("ecc counts unsupported here: %s" % e)
+    finally:
+        _amdsmi.shutdown()
+
+
+def test_amdsmi_event_notification_lifecycle():
+    """init → mask → poll(no events) → stop on the live driver."""
+    from kubevirt_gpu_device_plugin_amd import _amdsmi, amdsmi as smi
+    _amdsmi.init()
+    try:
+        try:
+            _amdsmi.event_init(0)
+        except RuntimeError as e:
+            pytest.skip("event notification unsupported: %s" % e)
+        _amdsmi.event_mask(0, smi.event_mask(
+            smi.EVT_GPU_PRE_RESET, smi.EVT_GPU_POST_RESET))
+        events = _amdsmi.get_events(100)
+        assert isinstance(events, list)
+        _amdsmi.event_stop(0)
+    finally:
+        _amdsmi.shutdown()
+
+
+def test_kfd_island_map_real_sysfs():
+    """Parse the real KFD topology; the GPU node must surface with a
+    plausible BDF that exists on the PCI bus."""
+    from kubevirt_gpu_device_plugin_amd.topology import (
+        island_map_from_kfd,
+    )
+    m = island_map_from_kfd()
+    if not m:
+        pytest.skip("no KFD GPU nodes visible (vfio-bound?)")
+    for bdf in m:
+        assert os.path.exists("/sys/bus/pci/devices/%s" % bdf), bdf
+
+
+def test_real_pci_walk_finds_amd_devices():
+    """The native scanner walks the real /sys/bus/pci/devices; on a GPU
+    box the MI355X is amdgpu-bound (not allocatable) but must appear in
+    a vendor-1002 scan."""
+    from kubevirt_gpu_device_plugin_amd import _sysfs
+    recs = _sysfs.scan_pci("/sys/bus/pci/devices", "1002")
+    gpus = [r for r in recs if r["driver"] == "amdgpu"]
+    assert len(gpus) >= 1, recs
+
+
+def test_smoke_entrypoint():
+    import __graft_entry__
+    __graft_entry__.smoke()
