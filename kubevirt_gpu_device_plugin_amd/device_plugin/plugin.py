@@ -99,82 +99,82 @@ class GenericDevicePlugin(DevicePluginBase):
                         host_path=host_path, container_path=host_path,
                         permissions=consts.DEVICE_PERMISSIONS))
 
+            # Resolve every requested BDF to its IOMMU group first; the
+            # whole request then revalidates in ONE native call (the
+            # reference re-reads sysfs per member per request,
+            # generic_device_plugin.go:383-410 — that is where its
+            # Allocate latency lives).
+            plan = []
             for bdf in req.devicesIDs:
                 group = self.registry.bdf_to_iommu.get(bdf)
                 members = self.registry.iommu_map.get(group, [])
-                if group is None or not members:
+                if group is None or not members \
+                        or not any(d.addr == bdf for d in members):
                     context.abort(
                         grpc.StatusCode.INVALID_ARGUMENT,
                         "invalid allocation request: unknown device: %s"
                         % bdf)
-                requested_found = any(d.addr == bdf for d in members)
-                if _sysfs is not None:
-                    # One C call revalidates the whole group and (for
-                    # iommufd) collects the cdev names.
-                    failed, vfio_devs = _sysfs.revalidate(
-                        base, [(d.addr, group) for d in members],
-                        consts.AMD_VENDOR_ID, iommufd)
-                    if failed:
-                        log.warning("revalidation failed for %s", failed)
+                plan.append((bdf, group, members))
+
+            if _sysfs is not None:
+                flat = [(d.addr, group)
+                        for _, group, members in plan for d in members]
+                failed, vfio_devs = _sysfs.revalidate(
+                    base, flat, consts.AMD_VENDOR_ID, iommufd)
+                if failed:
+                    log.warning("revalidation failed for %s", failed)
+                    context.abort(
+                        grpc.StatusCode.INVALID_ARGUMENT,
+                        "invalid allocation request: unknown "
+                        "device: %s" % failed)
+                cdevs = iter(vfio_devs)
+                for bdf, group, members in plan:
+                    if iommufd:
+                        for _ in members:
+                            add_spec(posixpath_join(
+                                self.config.vfio_dir, "devices",
+                                next(cdevs)))
+                    self._finish_bdf(bdf, group, env_devices, add_spec,
+                                     iommufd)
+                plan = []
+
+            for bdf, group, members in plan:  # pure-Python fallback
+                for dev in members:
+                    try:
+                        live_group = sysfs.read_link_basename(
+                            base, dev.addr, "iommu_group")
+                    except OSError:
+                        live_group = None
+                    if live_group != group:
+                        log.warning("IOMMU group changed for %s",
+                                    dev.addr)
                         context.abort(
                             grpc.StatusCode.INVALID_ARGUMENT,
                             "invalid allocation request: unknown "
-                            "device: %s" % failed)
+                            "device: %s" % dev.addr)
+                    try:
+                        vendor = sysfs.read_id_from_file(
+                            base, dev.addr, "vendor")
+                    except OSError:
+                        vendor = None
+                    if vendor != consts.AMD_VENDOR_ID:
+                        log.warning("vendor changed for %s", dev.addr)
+                        context.abort(
+                            grpc.StatusCode.INVALID_ARGUMENT,
+                            "invalid allocation request: unknown "
+                            "device: %s" % dev.addr)
                     if iommufd:
-                        for cdev in vfio_devs:
-                            add_spec(posixpath_join(
-                                self.config.vfio_dir, "devices", cdev))
-                else:
-                    for dev in members:
                         try:
-                            live_group = sysfs.read_link_basename(
-                                base, dev.addr, "iommu_group")
+                            vfiodev = sysfs.read_vfio_dev(base, dev.addr)
                         except OSError:
-                            live_group = None
-                        if live_group != group:
-                            log.warning("IOMMU group changed for %s",
-                                        dev.addr)
                             context.abort(
-                                grpc.StatusCode.INVALID_ARGUMENT,
-                                "invalid allocation request: unknown "
-                                "device: %s" % dev.addr)
-                        try:
-                            vendor = sysfs.read_id_from_file(
-                                base, dev.addr, "vendor")
-                        except OSError:
-                            vendor = None
-                        if vendor != consts.AMD_VENDOR_ID:
-                            log.warning("vendor changed for %s", dev.addr)
-                            context.abort(
-                                grpc.StatusCode.INVALID_ARGUMENT,
-                                "invalid allocation request: unknown "
-                                "device: %s" % dev.addr)
-                        if iommufd:
-                            try:
-                                vfiodev = sysfs.read_vfio_dev(
-                                    base, dev.addr)
-                            except OSError:
-                                context.abort(
-                                    grpc.StatusCode.INTERNAL,
-                                    "could not determine iommufd device "
-                                    "for device %s" % dev.addr)
-                            add_spec(posixpath_join(
-                                self.config.vfio_dir, "devices",
-                                vfiodev))
-                if not requested_found:
-                    context.abort(
-                        grpc.StatusCode.INVALID_ARGUMENT,
-                        "invalid allocation request: unknown device: %s"
-                        % bdf)
-
-                key = "%s_%s" % (self.env_prefix,
-                                 self.device_name.upper())
-                env_devices.setdefault(key, []).append(bdf)
-
-                add_spec(posixpath_join(self.config.vfio_dir, "vfio"))
-                add_spec(posixpath_join(self.config.vfio_dir, group))
-                if iommufd:
-                    add_spec(self.config.iommu_dev)
+                                grpc.StatusCode.INTERNAL,
+                                "could not determine iommufd device "
+                                "for device %s" % dev.addr)
+                        add_spec(posixpath_join(
+                            self.config.vfio_dir, "devices", vfiodev))
+                self._finish_bdf(bdf, group, env_devices, add_spec,
+                                 iommufd)
 
             container = response.container_responses.add()
             for key, bdfs in env_devices.items():
@@ -183,6 +183,16 @@ class GenericDevicePlugin(DevicePluginBase):
             log.info("[%s] allocated: envs=%s specs=%d", self.device_name,
                      dict(container.envs), len(specs))
         return response
+
+    def _finish_bdf(self, bdf, group, env_devices, add_spec, iommufd):
+        """Env entry (requested BDF only) + the per-group device specs,
+        in reference order (generic_device_plugin.go:414-432)."""
+        key = "%s_%s" % (self.env_prefix, self.device_name.upper())
+        env_devices.setdefault(key, []).append(bdf)
+        add_spec(posixpath_join(self.config.vfio_dir, "vfio"))
+        add_spec(posixpath_join(self.config.vfio_dir, group))
+        if iommufd:
+            add_spec(self.config.iommu_dev)
 
     # ---- GetPreferredAllocation ----------------------------------------
 
