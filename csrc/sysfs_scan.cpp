@@ -138,9 +138,11 @@ py::list scan_pci(const std::string &base_path, const std::string &vendor) {
 
 // TOCTOU revalidation for Allocate: for each (addr, expected_group),
 // confirm the iommu_group link and vendor are unchanged
-// (reference: generic_device_plugin.go:389-398).  Returns the first
-// failing addr or "" when all pass; optionally collects vfio-dev cdev
-// names (iommufd flow) into `vfio_devs`.
+// (reference: generic_device_plugin.go:389-398).  Returns
+// (failed_addr, reason, vfio_devs): failed_addr "" when all pass;
+// reason "changed" (group/vendor drift — invalid request) or "no_cdev"
+// (iommufd cdev missing — internal error, matching the reference's
+// error split, generic_device_plugin.go:403-409).
 py::tuple revalidate(const std::string &base_path,
                      const std::vector<std::pair<std::string, std::string>>
                          &addr_groups,
@@ -150,31 +152,38 @@ py::tuple revalidate(const std::string &base_path,
     return py::make_tuple(addr_groups.empty()
                               ? std::string("")
                               : addr_groups.front().first,
-                          py::list());
+                          std::string("changed"), py::list());
   py::list vfio_devs;
   for (const auto &ag : addr_groups) {
     int devfd = openat(base, ag.first.c_str(),
                        O_RDONLY | O_DIRECTORY | O_CLOEXEC);
     if (devfd < 0) {
       close(base);
-      return py::make_tuple(ag.first, py::list());
+      return py::make_tuple(ag.first, std::string("changed"),
+                            py::list());
     }
     std::string group, v;
     bool ok = link_basename(devfd, "iommu_group", &group) &&
               group == ag.second && read_id(devfd, "vendor", &v) &&
               v == vendor;
+    if (!ok) {
+      close(devfd);
+      close(base);
+      return py::make_tuple(ag.first, std::string("changed"),
+                            py::list());
+    }
     std::string cdev;
-    if (ok && want_vfio_dev) {
+    if (want_vfio_dev) {
+      bool found = false;
       int vd = openat(devfd, "vfio-dev",
                       O_RDONLY | O_DIRECTORY | O_CLOEXEC);
-      ok = false;
       if (vd >= 0) {
         DIR *vdir = fdopendir(vd);
         if (vdir) {
           while (struct dirent *de = readdir(vdir)) {
             if (strncmp(de->d_name, "vfio", 4) == 0) {
               cdev = de->d_name;
-              ok = true;
+              found = true;
               break;
             }
           }
@@ -183,16 +192,18 @@ py::tuple revalidate(const std::string &base_path,
           close(vd);
         }
       }
+      if (!found) {
+        close(devfd);
+        close(base);
+        return py::make_tuple(ag.first, std::string("no_cdev"),
+                              py::list());
+      }
+      vfio_devs.append(cdev);
     }
     close(devfd);
-    if (!ok) {
-      close(base);
-      return py::make_tuple(ag.first, py::list());
-    }
-    if (want_vfio_dev) vfio_devs.append(cdev);
   }
   close(base);
-  return py::make_tuple(std::string(""), vfio_devs);
+  return py::make_tuple(std::string(""), std::string(""), vfio_devs);
 }
 
 }  // namespace

@@ -119,8 +119,13 @@ class GenericDevicePlugin(DevicePluginBase):
             if _sysfs is not None:
                 flat = [(d.addr, group)
                         for _, group, members in plan for d in members]
-                failed, vfio_devs = _sysfs.revalidate(
+                failed, reason, vfio_devs = _sysfs.revalidate(
                     base, flat, consts.AMD_VENDOR_ID, iommufd)
+                if failed and reason == "no_cdev":
+                    context.abort(
+                        grpc.StatusCode.INTERNAL,
+                        "could not determine iommufd device for "
+                        "device %s" % failed)
                 if failed:
                     log.warning("revalidation failed for %s", failed)
                     context.abort(

@@ -298,3 +298,31 @@ def test_get_preferred_allocation_must_include_too_big(rig,
                 allocation_size=1)]))
     assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     ch.close()
+
+
+def test_allocate_iommufd_missing_cdev_internal_error(rig,
+                                                      synthetic_host):
+    """iommufd host but device has no vfio-dev cdev → INTERNAL
+    (reference: generic_device_plugin.go:403-409)."""
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")  # no vfio_dev dir
+    h.enable_iommufd()
+    _, plugin, _ = rig()
+    ch, stub = dial_plugin(plugin.socket_path)
+    with pytest.raises(grpc.RpcError) as exc:
+        stub.Allocate(dpapi.AllocateRequest(
+            container_requests=[dpapi.ContainerAllocateRequest(
+                devicesIDs=["0000:0c:00.0"])]))
+    assert exc.value.code() == grpc.StatusCode.INTERNAL
+    assert "iommufd" in exc.value.details()
+    ch.close()
+
+
+def test_pre_start_container(rig, synthetic_host):
+    synthetic_host.add_gpu("0000:0c:00.0", iommu_group="40")
+    _, plugin, _ = rig()
+    ch, stub = dial_plugin(plugin.socket_path)
+    resp = stub.PreStartContainer(dpapi.PreStartContainerRequest(
+        devicesIDs=["0000:0c:00.0"]))
+    assert resp == dpapi.PreStartContainerResponse()
+    ch.close()
