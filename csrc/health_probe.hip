@@ -34,27 +34,38 @@ namespace {
 
 // Pattern fill: each lane writes a value derived from its index so the
 // verify pass can detect addressing faults, not just stuck bits.
+// Non-temporal dwordx4 stores: the buffer is write-once-read-once, so
+// bypassing L2 keeps the stream at HBM3E rate instead of thrashing the
+// per-XCD L2.  Contiguous per-wavefront accesses (i = global thread id,
+// stride = grid) coalesce into full 4 KiB bursts.
 __global__ void fill_kernel(uint4 *__restrict__ out, size_t n_vec,
                             uint32_t seed) {
-  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  size_t stride = gridDim.x * blockDim.x;
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < n_vec; i += stride) {
     uint32_t base = seed ^ (uint32_t)(i * 2654435761u);
-    out[i] = make_uint4(base, base + 1, base + 2, base + 3);
+    uint4 v = make_uint4(base, base + 1, base + 2, base + 3);
+    __builtin_nontemporal_store(v.x, &out[i].x);
+    __builtin_nontemporal_store(v.y, &out[i].y);
+    __builtin_nontemporal_store(v.z, &out[i].z);
+    __builtin_nontemporal_store(v.w, &out[i].w);
   }
 }
 
-// Verify + bandwidth read pass: XOR-reduce everything; mismatches flip
-// the error counter.
+// Verify + bandwidth read pass: mismatches flip the error counter.
 __global__ void verify_kernel(const uint4 *__restrict__ in, size_t n_vec,
                               uint32_t seed,
                               unsigned long long *__restrict__ errors) {
-  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  size_t stride = gridDim.x * blockDim.x;
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
   unsigned long long local = 0;
   for (; i < n_vec; i += stride) {
     uint32_t base = seed ^ (uint32_t)(i * 2654435761u);
-    uint4 v = in[i];
+    uint4 v;
+    v.x = __builtin_nontemporal_load(&in[i].x);
+    v.y = __builtin_nontemporal_load(&in[i].y);
+    v.z = __builtin_nontemporal_load(&in[i].z);
+    v.w = __builtin_nontemporal_load(&in[i].w);
     local += (v.x != base) + (v.y != base + 1) + (v.z != base + 2) +
              (v.w != base + 3);
   }
