@@ -17,6 +17,7 @@ def main():
         format="%(asctime)s %(levelname).1s %(name)s: %(message)s")
     logging.info("starting AMD KubeVirt GPU device plugin")
     stop_event = threading.Event()
+    rescan_event = threading.Event()
 
     def _on_signal(signum, frame):
         logging.info("received signal %d; shutting down", signum)
@@ -24,7 +25,10 @@ def main():
 
     signal.signal(signal.SIGTERM, _on_signal)
     signal.signal(signal.SIGINT, _on_signal)
-    initiate_device_plugin(stop_event=stop_event)
+    # SIGHUP = re-discover (e.g. after `echo 8 > sriov_numvfs`)
+    signal.signal(signal.SIGHUP, lambda *a: rescan_event.set())
+    initiate_device_plugin(stop_event=stop_event,
+                           rescan_event=rescan_event)
 
 
 if __name__ == "__main__":

@@ -108,13 +108,29 @@ class Controller:
                 log.exception("error stopping %s", p.device_name)
 
 
-def initiate_device_plugin(stop_event=None, **kwargs):
+def initiate_device_plugin(stop_event=None, rescan_event=None, **kwargs):
     """Blocking entry point (reference: InitiateDevicePlugin,
-    device_plugin.go:89-96)."""
+    device_plugin.go:89-96).
+
+    ``rescan_event`` (set by SIGHUP in cmd/main.py) triggers a full
+    re-discovery + re-registration cycle.  The reference discovers
+    exactly once and needs a process restart to pick up new devices
+    (SURVEY.md §5 "no hotplug re-scan") — on MI355X that matters:
+    gim instantiates VFs *after* daemon start (``echo 8 >
+    sriov_numvfs``), so an operator can HUP the daemon instead of
+    bouncing it.
+    """
     stop_event = stop_event or threading.Event()
     controller = Controller(**kwargs)
     controller.create_plugins()
     controller.start(stop_event)
-    stop_event.wait()
+    while not stop_event.is_set():
+        stop_event.wait(0.5)
+        if rescan_event is not None and rescan_event.is_set():
+            rescan_event.clear()
+            log.info("rescan requested: re-running discovery")
+            controller.stop()
+            controller.create_plugins()
+            controller.start(stop_event)
     log.info("shutting down device plugin controller")
     controller.stop()

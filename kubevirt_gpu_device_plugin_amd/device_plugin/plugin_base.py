@@ -86,6 +86,19 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         self._server = server
 
         self._wait_for_ready()
+
+        # Arm the health watcher BEFORE registering: the moment kubelet
+        # knows the socket it may remove/recreate it, and a removal
+        # before the watcher is armed would be missed (the reference
+        # registers first, generic_device_plugin.go:246-252 — benign
+        # there only because fsnotify setup races the same way).
+        self._watch_armed = threading.Event()
+        self._health_thread = threading.Thread(
+            target=self._health_loop_guard,
+            name="health-%s" % self.device_name, daemon=True)
+        self._health_thread.start()
+        self._watch_armed.wait(timeout=2.0)
+
         try:
             self.register()
         except grpc.RpcError as e:
@@ -93,15 +106,6 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                       self.device_name, e)
             self.stop()
             raise
-
-        self._watch_armed = threading.Event()
-        self._health_thread = threading.Thread(
-            target=self._health_loop_guard,
-            name="health-%s" % self.device_name, daemon=True)
-        self._health_thread.start()
-        # Don't report ready until the watcher is armed, so a socket
-        # removal immediately after start is never missed.
-        self._watch_armed.wait(timeout=2.0)
         log.info("%s device plugin server ready", self.device_name)
 
     def stop(self):

@@ -65,6 +65,27 @@ class VfDevicePlugin(GenericDevicePlugin):
             name="smi-events-%s" % self.device_name, daemon=True)
         self._event_thread.start()
 
+    def Allocate(self, request, context):  # noqa: N802
+        """VF allocation = passthrough allocation, plus resource-type
+        validation: every requested BDF must be a VF of *this* type.
+
+        Deviation from the reference: its vGPU Allocate silently skips
+        ids whose mdev type mismatches, returning an empty env
+        (generic_vgpu_device_plugin.go:218-223); a loud INVALID_ARGUMENT
+        beats a VM that boots without its GPU.
+        """
+        import grpc
+
+        mine = {d.ID for d in self._devs}
+        for req in request.container_requests:
+            for bdf in req.devicesIDs:
+                if bdf not in mine:
+                    context.abort(
+                        grpc.StatusCode.INVALID_ARGUMENT,
+                        "invalid allocation request: %s is not a %s "
+                        "device" % (bdf, self.device_name))
+        return super().Allocate(request, context)
+
     def _vfs_of_pf(self, pf_bdf):
         """Child VFs of a PF that belong to *this* resource type."""
         mine = {d.ID for d in self._devs}

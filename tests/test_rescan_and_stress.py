@@ -1,0 +1,156 @@
+"""Rescan (hotplug) flow and concurrency/restart stress tests."""
+
+import threading
+from concurrent.futures import ThreadPoolExecutor
+
+from kubevirt_gpu_device_plugin_amd import dpapi
+from kubevirt_gpu_device_plugin_amd.device_plugin import discovery
+from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
+    Controller, build_kubelet_devices, initiate_device_plugin,
+)
+from kubevirt_gpu_device_plugin_amd.device_plugin.plugin import (
+    GenericDevicePlugin,
+)
+from tests.fixtures import StubKubelet, dial_plugin, eventually
+
+
+def test_rescan_picks_up_new_vfs(synthetic_host):
+    """gim creates VFs after daemon start → SIGHUP analogue (rescan
+    event) makes them allocatable without a process restart (the
+    reference cannot: SURVEY.md §5 'no hotplug re-scan')."""
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", iommu_group="100")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    stop = threading.Event()
+    rescan = threading.Event()
+    t = threading.Thread(
+        target=initiate_device_plugin,
+        kwargs=dict(stop_event=stop, rescan_event=rescan, config=cfg,
+                    kfd_nodes_dir=h.kfd_nodes,
+                    vf_event_watcher_factory=lambda: None),
+        daemon=True)
+    t.start()
+    try:
+        assert kubelet.wait_register(10).resource_name \
+            == "amd.com/INSTINCT_MI355X"
+        # VFs appear later (echo 8 > sriov_numvfs on a gim PF)
+        h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+        h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0",
+                 iommu_group="120")
+        rescan.set()
+        names = {kubelet.wait_register(10).resource_name
+                 for _ in range(2)}
+        assert names == {"amd.com/INSTINCT_MI355X",
+                         "amd.com/INSTINCT_MI355X_VF"}
+    finally:
+        stop.set()
+        t.join(timeout=10)
+        kubelet.stop()
+
+
+def test_concurrent_allocates(synthetic_host):
+    """Parallel Allocate RPCs (kubelet admits pods concurrently) must
+    not corrupt each other's specs/envs."""
+    h = synthetic_host
+    for g in range(8):
+        h.add_gpu("0000:%02x:00.0" % (0x10 + g),
+                  iommu_group=str(100 + g))
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X",
+        build_kubelet_devices(reg.device_map["75a3"]), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        ch, stub = dial_plugin(plugin.socket_path)
+        bdfs = sorted(reg.bdf_to_iommu)
+
+        def one(i):
+            bdf = bdfs[i % 8]
+            resp = stub.Allocate(dpapi.AllocateRequest(
+                container_requests=[dpapi.ContainerAllocateRequest(
+                    devicesIDs=[bdf])]))
+            c = resp.container_responses[0]
+            assert dict(c.envs) == {
+                "PCI_RESOURCE_AMD_COM_INSTINCT_MI355X": bdf}
+            assert [d.host_path for d in c.devices] == [
+                h.vfio_dir + "/vfio",
+                h.vfio_dir + "/" + reg.bdf_to_iommu[bdf]]
+            return True
+
+        with ThreadPoolExecutor(8) as ex:
+            assert all(ex.map(one, range(64)))
+        ch.close()
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
+
+
+def test_repeated_kubelet_restarts(synthetic_host):
+    """Several kubelet restarts in a row: plugin re-registers each time
+    and keeps serving."""
+    import os
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X",
+        build_kubelet_devices(reg.device_map["75a3"]), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        kubelet.wait_register(5)
+        for _ in range(3):
+            os.remove(plugin.socket_path)
+            req = kubelet.wait_register(10)
+            assert req.resource_name == "amd.com/INSTINCT_MI355X"
+            eventually(lambda: os.path.exists(plugin.socket_path))
+        ch, stub = dial_plugin(plugin.socket_path)
+        assert stub.GetDevicePluginOptions(
+            dpapi.Empty()).get_preferred_allocation_available
+        ch.close()
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
+
+
+def test_vf_type_validation(synthetic_host):
+    """A passthrough BDF sent to the VF plugin is rejected loudly."""
+    import grpc
+    import pytest
+    from kubevirt_gpu_device_plugin_amd.device_plugin.vf_plugin import (
+        VfDevicePlugin,
+    )
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", iommu_group="100")
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0", iommu_group="120")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = VfDevicePlugin(
+        "INSTINCT_MI355X_VF",
+        build_kubelet_devices(reg.vf_map["75b3"]), reg, config=cfg,
+        event_watcher_factory=lambda: None)
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        ch, stub = dial_plugin(plugin.socket_path)
+        with pytest.raises(grpc.RpcError) as exc:
+            stub.Allocate(dpapi.AllocateRequest(
+                container_requests=[dpapi.ContainerAllocateRequest(
+                    devicesIDs=["0000:10:00.0"])]))
+        assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+        assert "is not a INSTINCT_MI355X_VF" in exc.value.details()
+        ch.close()
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
