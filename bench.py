@@ -36,6 +36,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--iommufd", action="store_true",
                     help="bench the iommufd cdev flow")
+    ap.add_argument("--vfs-per-gpu", type=int, default=0,
+                    help="SR-IOV mode: N gim PFs x this many VFs; each "
+                         "RPC allocates one PF's VF set")
     ap.add_argument("--vf-check", action="store_true",
                     help="also verify the 64-VF SR-IOV config counts")
     args = ap.parse_args()
@@ -68,7 +71,8 @@ def main():
     if rank == 0:
         from bench_harness.rig import measure_allocate
         lat, n_devices = measure_allocate(
-            args.gpus, args.steps, args.warmup, iommufd=args.iommufd)
+            args.gpus, args.steps, args.warmup, iommufd=args.iommufd,
+            vfs_per_gpu=args.vfs_per_gpu)
     elapsed = time.perf_counter() - t_start
     barrier_sync()
 
@@ -110,7 +114,11 @@ def main():
             "config": {
                 "model": "kubevirt-gpu-device-plugin-amd",
                 "allocatable_gpus": n_devices,
-                "allocate_request": "all %d devices per RPC" % n_devices,
+                "allocate_request": (
+                    "%d VFs (one PF set) per RPC" % args.vfs_per_gpu
+                    if args.vfs_per_gpu
+                    else "all %d devices per RPC" % n_devices),
+                "vfs_per_gpu": args.vfs_per_gpu,
                 "iommufd": bool(args.iommufd),
                 "p99_us": round(p99_us, 1),
                 "vf_config_allocatable": vf_counts,

@@ -74,11 +74,16 @@ class PluginProcess:
 
 
 def measure_allocate(n_gpus, steps, warmup, iommufd=False,
-                     allocate_all=True):
+                     allocate_all=True, vfs_per_gpu=0):
     """Start the rig, wait for registration + full device list, then
-    time `steps` Allocate RPCs.  Returns (latencies_s, n_advertised)."""
+    time `steps` Allocate RPCs.  Returns (latencies_s, n_advertised).
+
+    ``vfs_per_gpu`` > 0 switches to the SR-IOV config (BASELINE config
+    4): n_gpus gim PFs × vfs_per_gpu VFs; the VF resource is the one
+    advertised, and each RPC allocates one PF's worth of VFs."""
     with tempfile.TemporaryDirectory() as tmp:
-        host = build_node(tmp, n_gpus, iommufd=iommufd)
+        host = build_node(tmp, n_gpus, vfs_per_gpu=vfs_per_gpu,
+                          iommufd=iommufd)
         cfg = host.config()
         kubelet = StubKubelet(cfg.kubelet_socket)
         plugin = PluginProcess(host)
@@ -88,10 +93,20 @@ def measure_allocate(n_gpus, steps, warmup, iommufd=False,
             ch, stub = dial_plugin(sock, timeout=10.0)
             stream = stub.ListAndWatch(dpapi.Empty())
             devices = [d.ID for d in next(stream).devices]
-            assert len(devices) == n_gpus, \
-                "advertised %d != %d" % (len(devices), n_gpus)
+            expected = n_gpus * vfs_per_gpu if vfs_per_gpu else n_gpus
+            assert len(devices) == expected, \
+                "advertised %d != %d" % (len(devices), expected)
 
-            if allocate_all:
+            if vfs_per_gpu:
+                # one full GPU's VF set per RPC, round-robin over PFs
+                by_pf = {}
+                for d in devices:
+                    by_pf.setdefault(d.rsplit(".", 1)[0][:7], []) \
+                        .append(d)
+                reqs = [dpapi.AllocateRequest(container_requests=[
+                    dpapi.ContainerAllocateRequest(devicesIDs=vfs)])
+                    for vfs in by_pf.values()]
+            elif allocate_all:
                 reqs = [dpapi.AllocateRequest(container_requests=[
                     dpapi.ContainerAllocateRequest(devicesIDs=devices)])]
             else:
