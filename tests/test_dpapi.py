@@ -73,3 +73,27 @@ def test_roundtrip_device_topology():
         nodes=[dpapi.NUMANode(ID=0)]))
     d4 = dpapi.Device.FromString(d3.SerializeToString())
     assert len(d4.topology.nodes) == 1 and d4.topology.nodes[0].ID == 0
+
+
+def test_golden_wire_bytes():
+    """Pin exact wire encodings (hand-verified against the proto3 wire
+    format): field 1 LEN "v1beta1", etc.  Any drift in field numbers or
+    types breaks kubelet compatibility and must fail loudly here."""
+    r = dpapi.RegisterRequest(version="v1beta1",
+                              endpoint="kubevirt-X.sock",
+                              resource_name="amd.com/X")
+    assert r.SerializeToString().hex() == (
+        "0a0776316265746131120f6b756265766972742d582e736f636b"
+        "1a09616d642e636f6d2f58")
+    d = dpapi.Device(ID="0000:0c:00.0", health="Healthy",
+                     topology=dpapi.TopologyInfo(
+                         nodes=[dpapi.NUMANode(ID=1)]))
+    assert d.SerializeToString().hex() == (
+        "0a0c303030303a30633a30302e3012074865616c7468791a040a020801")
+    a = dpapi.ContainerAllocateResponse()
+    a.envs["K"] = "V"
+    a.devices.add(container_path="/dev/vfio/vfio",
+                  host_path="/dev/vfio/vfio", permissions="mrw")
+    assert a.SerializeToString().hex() == (
+        "0a060a014b1201561a250a0e2f6465762f7666696f2f7666696f"
+        "120e2f6465762f7666696f2f7666696f1a036d7277")
