@@ -59,17 +59,21 @@ def _lookup_in_file(path, device_id, vendor_id):
 
 
 def get_device_name(device_id, vendor_id=consts.AMD_VENDOR_ID,
-                    pci_ids_path=None):
+                    pci_ids_path=None, search_paths=None):
     """Resolve a sanitized device name for ``device_id``.
 
-    Search order: explicit ``pci_ids_path`` → the system database at
-    ``/usr/pci.ids`` (full db, embedded by the container build) → the
-    curated built-in table.  Returns "" when not found anywhere; the
-    controller then falls back to the raw hex id
-    (reference: device_plugin.go:124-128).
+    Search order: explicit ``pci_ids_path`` → ``search_paths``
+    (default: the system database at ``/usr/pci.ids`` — the full db,
+    embedded by the container build — then the curated built-in table).
+    Returns "" when not found anywhere; the controller then falls back
+    to the raw hex id (reference: device_plugin.go:124-128).
     """
-    candidates = ([pci_ids_path] if pci_ids_path else
-                  [consts.PCI_IDS_FILE_PATH, BUILTIN_IDS_PATH])
+    if pci_ids_path:
+        candidates = [pci_ids_path]
+    elif search_paths is not None:
+        candidates = list(search_paths)
+    else:
+        candidates = [consts.PCI_IDS_FILE_PATH, BUILTIN_IDS_PATH]
     for path in candidates:
         try:
             name = _lookup_in_file(path, device_id, vendor_id)

@@ -85,3 +85,26 @@ def test_builtin_table_has_mi355x():
     assert pciids.get_device_name(
         "74a1", pci_ids_path=pciids.BUILTIN_IDS_PATH) \
         == "AQUA_VANJARAM_INSTINCT_MI300X"
+
+
+def test_search_path_order(tmp_path):
+    """The system db (first path) wins over the built-in table; the
+    built-in backfills ids the system db lacks."""
+    system = _ids_file(tmp_path, """\
+        1002  AMD
+        \t74a1  System Db Name For MI300X
+        """)
+    got = pciids.get_device_name(
+        "74a1", search_paths=[system, pciids.BUILTIN_IDS_PATH])
+    assert got == "SYSTEM_DB_NAME_FOR_MI300X"
+    # 75a3 absent from the fake system db → built-in fallback
+    got = pciids.get_device_name(
+        "75a3", search_paths=[system, pciids.BUILTIN_IDS_PATH])
+    assert got == "INSTINCT_MI355X"
+
+
+def test_missing_system_db_skipped(tmp_path):
+    got = pciids.get_device_name(
+        "75a3", search_paths=[str(tmp_path / "nope"),
+                              pciids.BUILTIN_IDS_PATH])
+    assert got == "INSTINCT_MI355X"
