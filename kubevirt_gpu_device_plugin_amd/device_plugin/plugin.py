@@ -31,10 +31,6 @@ except ImportError:
 log = logging.getLogger(__name__)
 
 
-def posixpath_join(*parts):
-    return "/".join(p.rstrip("/") for p in parts[:-1]) + "/" + parts[-1]
-
-
 class GenericDevicePlugin(DevicePluginBase):
     def __init__(self, device_name, devices, registry, config=None,
                  island_of=None, env_prefix=consts.GPU_ENV_PREFIX):
@@ -42,6 +38,23 @@ class GenericDevicePlugin(DevicePluginBase):
         self.registry = registry
         self.island_of = island_of or (lambda bdf: -1)
         self.env_prefix = env_prefix
+        # Allocate hot-path caches: the env key and every DeviceSpec
+        # host path are deterministic per plugin, so build them once
+        # (protobuf message construction dominates the handler
+        # otherwise; kubelet copies on extend(), reuse is safe).
+        self._env_key = "%s_%s" % (env_prefix, device_name.upper())
+        self._vfio_prefix = self.config.vfio_dir.rstrip("/") + "/"
+        self._cdev_prefix = self._vfio_prefix + "devices/"
+        self._spec_cache = {}
+
+    def _spec(self, host_path):
+        s = self._spec_cache.get(host_path)
+        if s is None:
+            s = dpapi.DeviceSpec(host_path=host_path,
+                                 container_path=host_path,
+                                 permissions=consts.DEVICE_PERMISSIONS)
+            self._spec_cache[host_path] = s
+        return s
 
     # ---- health wiring --------------------------------------------------
 
@@ -91,13 +104,12 @@ class GenericDevicePlugin(DevicePluginBase):
             specs = []
             seen = set()
             env_devices = {}
+            spec_of = self._spec
 
             def add_spec(host_path):
                 if host_path not in seen:
                     seen.add(host_path)
-                    specs.append(dpapi.DeviceSpec(
-                        host_path=host_path, container_path=host_path,
-                        permissions=consts.DEVICE_PERMISSIONS))
+                    specs.append(spec_of(host_path))
 
             # Resolve every requested BDF to its IOMMU group first; the
             # whole request then revalidates in ONE native call (the
@@ -136,9 +148,7 @@ class GenericDevicePlugin(DevicePluginBase):
                 for bdf, group, members in plan:
                     if iommufd:
                         for _ in members:
-                            add_spec(posixpath_join(
-                                self.config.vfio_dir, "devices",
-                                next(cdevs)))
+                            add_spec(self._cdev_prefix + next(cdevs))
                     self._finish_bdf(bdf, group, env_devices, add_spec,
                                      iommufd)
                 plan = []
@@ -176,8 +186,7 @@ class GenericDevicePlugin(DevicePluginBase):
                                 grpc.StatusCode.INTERNAL,
                                 "could not determine iommufd device "
                                 "for device %s" % dev.addr)
-                        add_spec(posixpath_join(
-                            self.config.vfio_dir, "devices", vfiodev))
+                        add_spec(self._cdev_prefix + vfiodev)
                 self._finish_bdf(bdf, group, env_devices, add_spec,
                                  iommufd)
 
@@ -192,10 +201,9 @@ class GenericDevicePlugin(DevicePluginBase):
     def _finish_bdf(self, bdf, group, env_devices, add_spec, iommufd):
         """Env entry (requested BDF only) + the per-group device specs,
         in reference order (generic_device_plugin.go:414-432)."""
-        key = "%s_%s" % (self.env_prefix, self.device_name.upper())
-        env_devices.setdefault(key, []).append(bdf)
-        add_spec(posixpath_join(self.config.vfio_dir, "vfio"))
-        add_spec(posixpath_join(self.config.vfio_dir, group))
+        env_devices.setdefault(self._env_key, []).append(bdf)
+        add_spec(self._vfio_prefix + "vfio")
+        add_spec(self._vfio_prefix + group)
         if iommufd:
             add_spec(self.config.iommu_dev)
 
