@@ -1,12 +1,14 @@
 // _healthprobe — gfx950 GPU self-test kernel.
 //
-// The plugin's host-side GPU sanity check for nodes where a PF is
-// host-driver-resident (SR-IOV): before VFs of a GPU are advertised
-// Healthy, a quick HBM write/readback at bandwidth plus an MFMA-unit
-// exercise confirms the silicon answers.  The reference has no GPU
-// compute at all (SURVEY.md §2 — its only native code is the NVML
-// binding); this is the MI355X-native extra that makes health checking
-// mean more than "the /dev node exists".
+// The plugin's GPU self-test for nodes where a GPU is
+// host-driver-resident (SR-IOV PFs): an HBM write/readback at bandwidth
+// plus an MFMA-unit exercise confirms the silicon answers.  Invoked by
+// operators (`make test-gpu`) and by the driver's smoke(); the daemon's
+// in-band health signals stay event-driven (vfio nodes + AMD-SMI).
+// The reference has no GPU compute at all (SURVEY.md §2 — its only
+// native code is the NVML binding); this is the MI355X-native extra
+// that makes "is this GPU healthy" answerable beyond "the /dev node
+// exists".
 //
 // CDNA4 notes (per /opt/skills/guides/MI355X_MICROARCH.md): 64-wide
 // wavefronts, 256 threads/block, grid-stride loops sized ≫256

@@ -212,8 +212,10 @@ class GenericDevicePlugin(DevicePluginBase):
     def GetPreferredAllocation(self, request, context):  # noqa: N802
         """NUMA + xGMI-island preferred sets (reference is NUMA-only,
         generic_device_plugin.go:478-616; see allocation.py)."""
-        numa = {d.ID: (d.topology.nodes[0].ID if d.topology.nodes else -1)
-                for d in self.devices_snapshot()}
+        with self._lock:
+            numa = {d.ID: (d.topology.nodes[0].ID if d.topology.nodes
+                           else -1)
+                    for d in self._devs}
         response = dpapi.PreferredAllocationResponse()
         for req in request.container_requests:
             try:
