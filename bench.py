@@ -51,7 +51,10 @@ def main():
         import torch
         import torch.distributed as tdist
         dist = tdist
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL only when every rank actually has its own GPU
+        backend = ("nccl" if torch.cuda.is_available()
+                   and torch.cuda.device_count() >= world_size
+                   else "gloo")
         if backend == "nccl":
             local_rank = int(os.environ.get("LOCAL_RANK", rank))
             device = torch.device("cuda", local_rank)

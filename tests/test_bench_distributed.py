@@ -10,12 +10,14 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_bench_under_torchrun_gloo():
+    env = dict(os.environ)
+    env["CUDA_VISIBLE_DEVICES"] = ""  # force gloo even on a GPU box
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
          "--master-port", "29597", "bench.py", "--gpus", "2",
          "--steps", "5", "--warmup", "1"],
-        cwd=REPO, capture_output=True, text=True, timeout=240)
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=240)
     assert out.returncode == 0, out.stderr[-2000:]
     line = [l for l in out.stdout.splitlines()
             if l.startswith("{")][-1]
