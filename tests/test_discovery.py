@@ -125,3 +125,13 @@ def test_native_python_parity(synthetic_host):
         == {k: [(d.addr, d.numa_node) for d in v]
             for k, v in reg_py.device_map.items()}
     assert reg_native.pf_vf_map == reg_py.pf_vf_map
+
+
+def test_garbage_numa_defaults_to_zero(synthetic_host):
+    import os
+    h = synthetic_host
+    d = h.add_gpu("0000:0c:00.0")
+    with open(os.path.join(d, "numa_node"), "w") as f:
+        f.write("not-a-number\n")
+    reg = discovery.discover(base_path=h.pci, use_native=False)
+    assert reg.device_map["75a3"][0].numa_node == 0

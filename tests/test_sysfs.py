@@ -64,3 +64,13 @@ def test_read_vfio_dev_missing(synthetic_host):
     h.add_gpu("0000:0c:00.0")
     with pytest.raises(OSError):
         sysfs.read_vfio_dev(h.pci, "0000:0c:00.0")
+
+
+def test_read_sriov_numvfs(synthetic_host):
+    import os
+    h = synthetic_host
+    d = h.add_gpu("0000:0c:00.0")
+    assert sysfs.read_sriov_numvfs(h.pci, "0000:0c:00.0") == 0
+    with open(os.path.join(d, "sriov_numvfs"), "w") as f:
+        f.write("8\n")
+    assert sysfs.read_sriov_numvfs(h.pci, "0000:0c:00.0") == 8
