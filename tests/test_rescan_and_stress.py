@@ -154,3 +154,68 @@ def test_vf_type_validation(synthetic_host):
         stop.set()
         plugin.stop()
         kubelet.stop()
+
+
+def test_no_fd_leak_across_restarts(synthetic_host):
+    """Each restart churns a gRPC server, an inotify fd and a kubelet
+    channel; fd count must stay stable."""
+    import os
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X",
+        build_kubelet_devices(reg.device_map["75a3"]), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        kubelet.wait_register(5)
+        plugin.restart()
+        kubelet.wait_register(5)
+        before = len(os.listdir("/proc/self/fd"))
+        for _ in range(5):
+            plugin.restart()
+            kubelet.wait_register(5)
+        after = len(os.listdir("/proc/self/fd"))
+        assert after - before <= 4, (before, after)
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
+
+
+def test_vf_plugin_with_real_smi_binding_degrades(synthetic_host):
+    """The REAL _amdsmi binding on a driverless host: init fails inside
+    the watcher thread, is logged, and the plugin keeps serving
+    (reference: generic_vgpu_device_plugin.go:290-297)."""
+    import pytest
+    from kubevirt_gpu_device_plugin_amd import amdsmi
+    from kubevirt_gpu_device_plugin_amd.device_plugin.vf_plugin import (
+        VfDevicePlugin,
+    )
+    if not amdsmi.is_available():
+        pytest.skip("_amdsmi extension not built")
+    h = synthetic_host
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0", iommu_group="120")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = VfDevicePlugin(
+        "INSTINCT_MI355X_VF",
+        build_kubelet_devices(reg.vf_map["75b3"]), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)  # default watcher factory → real binding
+    try:
+        ch, stub = dial_plugin(plugin.socket_path)
+        resp = stub.Allocate(dpapi.AllocateRequest(
+            container_requests=[dpapi.ContainerAllocateRequest(
+                devicesIDs=["0000:20:02.0"])]))
+        assert resp.container_responses[0].envs
+        ch.close()
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
