@@ -34,11 +34,35 @@ WATCH_MASK = event_mask(EVT_VMFAULT, EVT_THERMAL_THROTTLE,
 
 
 class AmdSmiEventWatcher:
-    """Blocking watch loop; run on a daemon thread by VfDevicePlugin."""
+    """Blocking watch loop; run on a daemon thread by VfDevicePlugin.
+
+    Two fault sources, both fanned out PF → child VFs:
+      * event notifications (reset/thermal/vmfault), handled per the
+        criticality table above;
+      * polled RAS totals: a growing *uncorrectable* ECC count marks
+        the GPU unhealthy (amdsmi has no RAS entry in the notification
+        enum, amdsmi.h:1336-1352, so this must be polled — the AMD
+        analogue of memory-related critical XIDs).
+    """
 
     def __init__(self, smi=None, poll_ms=POLL_MS):
         self._smi = smi or ext()
         self._poll_ms = poll_ms
+
+    def _check_ecc(self, smi, by_index, baseline, on_unhealthy):
+        for idx, bdf in by_index.items():
+            try:
+                ec = smi.ecc_count(idx)
+            except (RuntimeError, AttributeError):
+                continue  # RAS not supported here
+            bad = ec.get("uncorrectable", 0)
+            if idx not in baseline:
+                baseline[idx] = bad
+            elif bad > baseline[idx]:
+                log.warning("uncorrectable ECC errors on %s: %d "
+                            "(was %d)", bdf, bad, baseline[idx])
+                baseline[idx] = bad
+                on_unhealthy(bdf)
 
     def watch(self, pf_bdfs, on_unhealthy, on_healthy, should_stop):
         smi = self._smi
@@ -57,6 +81,8 @@ class AmdSmiEventWatcher:
             for idx in by_index:
                 smi.event_init(idx)
                 smi.event_mask(idx, WATCH_MASK)
+            ecc_baseline = {}
+            self._check_ecc(smi, by_index, ecc_baseline, on_unhealthy)
             try:
                 while not should_stop():
                     for idx, etype, msg in smi.get_events(self._poll_ms):
@@ -74,6 +100,8 @@ class AmdSmiEventWatcher:
                         else:
                             log.info("GPU event %d on %s: %s",
                                      etype, bdf, msg)
+                    self._check_ecc(smi, by_index, ecc_baseline,
+                                    on_unhealthy)
             finally:
                 for idx in by_index:
                     smi.event_stop(idx)

@@ -61,6 +61,9 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         self._term = threading.Event()
         self._watch_armed = threading.Event()
         self._health_thread = None
+        # serializes start/stop/restart between the controller thread
+        # and the health thread's kubelet-restart path
+        self._lifecycle = threading.RLock()
 
     # ---- lifecycle ------------------------------------------------------
 
@@ -71,6 +74,10 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
     def start(self, stop_event):
         """Create the socket, serve, self-dial, register with kubelet and
         start health watching (reference: Start, generic_device_plugin.go:217-257)."""
+        with self._lifecycle:
+            self._start_locked(stop_event)
+
+    def _start_locked(self, stop_event):
         if self._server is not None:
             raise RuntimeError("gRPC server already started")
         self._stop = stop_event
@@ -111,22 +118,30 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         log.info("%s device plugin server ready", self.device_name)
 
     def stop(self):
-        if self._server is None:
-            return
-        self._term.set()
-        with self._lock:
-            self._lock.notify_all()
-        server, self._server = self._server, None
-        server.stop(grace=None)
-        self._cleanup_socket()
+        with self._lifecycle:
+            if self._server is None:
+                return
+            self._term.set()
+            with self._lock:
+                self._lock.notify_all()
+            server, self._server = self._server, None
+            server.stop(grace=None)
+            self._cleanup_socket()
 
     def restart(self):
         """Full re-handshake after a kubelet restart
         (reference: restart, generic_device_plugin.go:275-286)."""
-        log.info("restarting %s device plugin server", self.device_name)
-        stop_event = self._stop
-        self.stop()
-        self.start(stop_event)
+        with self._lifecycle:
+            if self._should_exit():
+                # the daemon is shutting down concurrently — a restart
+                # here would resurrect a server the controller just
+                # stopped
+                return
+            log.info("restarting %s device plugin server",
+                     self.device_name)
+            stop_event = self._stop
+            self.stop()
+            self.start(stop_event)
 
     def _cleanup_socket(self):
         try:

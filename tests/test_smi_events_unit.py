@@ -72,3 +72,27 @@ def test_shutdown_called_after_watch():
     t.join(timeout=3.0)
     assert not smi.inited  # shutdown ran
     assert not smi.event_inited  # event_stop ran
+
+
+def test_ecc_uncorrectable_growth_marks_unhealthy():
+    smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u"}])
+    smi.ecc = {"correctable": 0, "uncorrectable": 0, "deferred": 0}
+    smi.ecc_count = lambda idx: dict(smi.ecc)
+    unhealthy, healthy, t = run_watch(smi, ["0000:0c:00.0"],
+                                      duration=2.0)
+    eventually(lambda: smi.event_inited == {0})
+    assert unhealthy == []  # baseline taken, no flip
+    smi.ecc["uncorrectable"] = 3
+    eventually(lambda: unhealthy == ["0000:0c:00.0"])
+    t.join(timeout=4.0)
+
+
+def test_ecc_unsupported_is_fine():
+    smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u"}])
+    def boom(idx):
+        raise RuntimeError("not supported")
+    smi.ecc_count = boom
+    unhealthy, healthy, t = run_watch(smi, ["0000:0c:00.0"],
+                                      duration=0.3)
+    t.join(timeout=3.0)
+    assert unhealthy == []
