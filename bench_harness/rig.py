@@ -127,3 +127,4 @@ def measure_allocate(n_gpus, steps, warmup, iommufd=False,
         finally:
             plugin.stop()
             kubelet.stop()
+            host.cleanup()

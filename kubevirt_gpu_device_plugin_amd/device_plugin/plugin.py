@@ -14,7 +14,6 @@ equivalent here.
 """
 
 import logging
-import os
 
 import grpc
 
@@ -95,9 +94,8 @@ class GenericDevicePlugin(DevicePluginBase):
         into container B's env).
         """
         base = self.config.pci_base
-        # /dev/iommu present ⇒ iommufd cdev flow (reference:
-        # supportsIOMMUFD, generic_device_plugin.go:700-709).
-        iommufd = os.path.exists(self.config.iommu_dev)
+        # /dev/iommu present ⇒ iommufd cdev flow
+        iommufd = sysfs.supports_iommufd(self.config.iommu_dev)
 
         response = dpapi.AllocateResponse()
         for req in request.container_requests:

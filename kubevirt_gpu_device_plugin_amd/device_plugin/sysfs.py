@@ -75,7 +75,7 @@ def read_vfio_dev(base_path, device_address):
                             % device_address)
 
 
-def supports_iommufd(root_path="/"):
+def supports_iommufd(iommu_dev_path="/dev/iommu"):
     """Host supports iommufd when /dev/iommu exists
     (reference: supportsIOMMUFD, generic_device_plugin.go:700-709)."""
-    return os.path.exists(os.path.join(root_path, "dev", "iommu"))
+    return os.path.exists(iommu_dev_path)

@@ -14,4 +14,6 @@ def pytest_configure(config):
 @pytest.fixture
 def synthetic_host(tmp_path):
     from tests.fixtures import SyntheticHost
-    return SyntheticHost(tmp_path)
+    h = SyntheticHost(tmp_path)
+    yield h
+    h.cleanup()

@@ -26,17 +26,24 @@ class SyntheticHost:
     """Builds a fake host filesystem under a tempdir."""
 
     def __init__(self, root):
+        import tempfile
         self.root = str(root)
         self.pci = os.path.join(self.root, "sys", "bus", "pci", "devices")
         self.vfio_dir = os.path.join(self.root, "dev", "vfio")
         self.iommu_dev = os.path.join(self.root, "dev", "iommu")
-        self.kubelet_dir = os.path.join(self.root, "kubelet")
+        # unix sockets live here — must stay SHORT (AF_UNIX caps paths
+        # at ~107 bytes and pytest/xdist tmp paths exceed it)
+        self.kubelet_dir = tempfile.mkdtemp(prefix="dpk-")
         self.kfd_nodes = os.path.join(self.root, "kfd", "nodes")
         for d in (self.pci, self.vfio_dir, self.kubelet_dir,
                   self.kfd_nodes):
             os.makedirs(d, exist_ok=True)
         # the vfio container node always exists on a vfio host
         self._touch(os.path.join(self.vfio_dir, "vfio"))
+
+    def cleanup(self):
+        import shutil
+        shutil.rmtree(self.kubelet_dir, ignore_errors=True)
 
     @staticmethod
     def _touch(path):
