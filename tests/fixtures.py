@@ -11,7 +11,6 @@ register → dial-back → Allocate flow.
 
 import os
 import queue
-import threading
 from concurrent import futures
 
 import grpc

@@ -6,7 +6,7 @@ from concurrent.futures import ThreadPoolExecutor
 from kubevirt_gpu_device_plugin_amd import dpapi
 from kubevirt_gpu_device_plugin_amd.device_plugin import discovery
 from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
-    Controller, build_kubelet_devices, initiate_device_plugin,
+    build_kubelet_devices, initiate_device_plugin,
 )
 from kubevirt_gpu_device_plugin_amd.device_plugin.plugin import (
     GenericDevicePlugin,
