@@ -74,6 +74,32 @@ __global__ void verify_kernel(const uint4 *__restrict__ in, size_t n_vec,
   if (local) atomicAdd(errors, local);
 }
 
+// LDS exercise: every workgroup fills its 64 KB LDS slab with an
+// index-derived pattern, cross-reads it lane-swizzled (hits all 32
+// banks), and counts mismatches.  Launched with ≫256 workgroups so
+// every CU's LDS gets touched.
+__global__ void lds_kernel(unsigned long long *__restrict__ errors,
+                           uint32_t seed) {
+  __shared__ uint32_t slab[16384];  // 64 KB
+  unsigned t = threadIdx.x;
+  for (unsigned i = t; i < 16384; i += blockDim.x)
+    slab[i] = seed ^ (i * 2246822519u) ^ blockIdx.x;
+  __syncthreads();
+  unsigned long long local = 0;
+  for (unsigned i = t; i < 16384; i += blockDim.x) {
+    unsigned j = (i * 33) & 16383;  // stride-33 → bank-sweeping reads
+    if (slab[j] != (seed ^ (j * 2246822519u) ^ blockIdx.x)) ++local;
+  }
+  if (local) atomicAdd(errors, local);
+}
+
+// Device-scope atomics: every thread of every workgroup (spread across
+// all 8 XCDs) increments one counter; the total must be exact — a
+// quick cross-XCD coherency check.
+__global__ void atomic_kernel(unsigned long long *__restrict__ counter) {
+  atomicAdd(counter, 1ull);
+}
+
 // Minimal matrix-core exercise: one MFMA per wavefront, result checked
 // on host.  Confirms the XCD compute path beyond plain VALU/HBM.
 __global__ void mfma_kernel(float *__restrict__ out) {
@@ -148,6 +174,27 @@ py::dict probe(int device, size_t mib) {
   HIP_CHECK(hipMemcpy(&h_errors, errors, sizeof h_errors,
                       hipMemcpyDeviceToHost));
 
+  // LDS banks on every CU
+  HIP_CHECK(hipMemset(errors, 0, sizeof(unsigned long long)));
+  hipLaunchKernelGGL(lds_kernel, grid, block, 0, 0, errors, seed);
+  HIP_CHECK(hipDeviceSynchronize());
+  unsigned long long lds_errors = 0;
+  HIP_CHECK(hipMemcpy(&lds_errors, errors, sizeof lds_errors,
+                      hipMemcpyDeviceToHost));
+
+  // device-scope atomic coherency across XCDs
+  unsigned long long *counter = nullptr;
+  HIP_CHECK(hipMalloc(&counter, sizeof(unsigned long long)));
+  HIP_CHECK(hipMemset(counter, 0, sizeof(unsigned long long)));
+  hipLaunchKernelGGL(atomic_kernel, grid, block, 0, 0, counter);
+  HIP_CHECK(hipDeviceSynchronize());
+  unsigned long long h_counter = 0;
+  HIP_CHECK(hipMemcpy(&h_counter, counter, sizeof h_counter,
+                      hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(counter));
+  bool atomics_ok =
+      h_counter == (unsigned long long)blocks * block.x;
+
   float *mfma_out = nullptr;
   HIP_CHECK(hipMalloc(&mfma_out, 256 * sizeof(float)));
   HIP_CHECK(hipMemset(mfma_out, 0, 256 * sizeof(float)));
@@ -163,17 +210,20 @@ py::dict probe(int device, size_t mib) {
   HIP_CHECK(hipFree(mfma_out));
   HIP_CHECK(hipFree(errors));
   HIP_CHECK(hipFree(buf));
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
-  hipEventDestroy(t2);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  (void)hipEventDestroy(t2);
 
   double gib = static_cast<double>(bytes) / (1 << 30);
   result["bytes"] = bytes;
   result["write_gbps"] = gib / (fill_ms / 1e3);
   result["read_gbps"] = gib / (verify_ms / 1e3);
   result["pattern_errors"] = h_errors;
+  result["lds_errors"] = lds_errors;
+  result["atomics_ok"] = atomics_ok;
   result["mfma_ok"] = mfma_ok;
-  result["ok"] = (h_errors == 0) && mfma_ok;
+  result["ok"] = (h_errors == 0) && (lds_errors == 0) && atomics_ok &&
+                 mfma_ok;
   return result;
 }
 

@@ -42,6 +42,8 @@ def test_health_probe_kernel():
     from kubevirt_gpu_device_plugin_amd import _healthprobe
     r = _healthprobe.probe(0, 1024)
     assert r["pattern_errors"] == 0
+    assert r["lds_errors"] == 0
+    assert r["atomics_ok"]
     assert r["mfma_ok"], "MFMA 16x16x4 f32 result mismatch"
     assert r["ok"]
     assert "gfx950" in r["gcn_arch"], r["gcn_arch"]
