@@ -212,21 +212,26 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
 
     def ListAndWatch(self, request, context):  # noqa: N802
         """Initial full list, then a fresh full list on every health
-        transition (reference: generic_device_plugin.go:313-350)."""
+        transition (reference: generic_device_plugin.go:313-350).
+
+        Responses are built under the lock but yielded outside it —
+        a slow kubelet consumer must never block the health producers
+        or other streams.
+        """
         log.info("[%s] ListAndWatch: sending %d devices",
                  self.device_name, len(self._devs))
-        with self._lock:
-            last = self._version
-            yield dpapi.ListAndWatchResponse(devices=self._devs)
+        last = None
         while context.is_active():
             with self._lock:
-                if self._version == last and not self._should_exit():
+                while last == self._version and not self._should_exit():
                     self._lock.wait(timeout=0.5)
+                    if not context.is_active():
+                        return
                 if self._should_exit():
                     return
-                if self._version != last:
-                    last = self._version
-                    yield dpapi.ListAndWatchResponse(devices=self._devs)
+                last = self._version
+                resp = dpapi.ListAndWatchResponse(devices=self._devs)
+            yield resp
 
     def _should_exit(self):
         return self._term.is_set() or (self._stop is not None
