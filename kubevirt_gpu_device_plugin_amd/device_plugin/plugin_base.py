@@ -20,6 +20,7 @@ Differences from the reference, by design:
 import logging
 import os
 import threading
+import time
 from concurrent import futures
 from dataclasses import dataclass
 
@@ -130,7 +131,15 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
 
     def restart(self):
         """Full re-handshake after a kubelet restart
-        (reference: restart, generic_device_plugin.go:275-286)."""
+        (reference: restart, generic_device_plugin.go:275-286).
+
+        Registration is retried with backoff: a restarting kubelet
+        removes the plugin socket *before* its Registration service is
+        back up, so the first re-register attempt routinely races it.
+        (The reference gives up after one failed attempt and the plugin
+        stays dead until the daemon is bounced,
+        generic_device_plugin.go:688-692.)
+        """
         with self._lifecycle:
             if self._should_exit():
                 # the daemon is shutting down concurrently — a restart
@@ -141,7 +150,25 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                      self.device_name)
             stop_event = self._stop
             self.stop()
-            self.start(stop_event)
+            for attempt in range(30):
+                if stop_event is not None and stop_event.is_set():
+                    return
+                try:
+                    self.start(stop_event)
+                    return
+                except (grpc.RpcError, grpc.FutureTimeoutError,
+                        OSError) as e:
+                    log.warning(
+                        "[%s] restart attempt %d failed (%s); kubelet "
+                        "may still be coming up — retrying",
+                        self.device_name, attempt + 1, e)
+                    self.stop()
+                    if stop_event is not None:
+                        stop_event.wait(2.0)  # wakes early on shutdown
+                    else:
+                        time.sleep(2.0)
+            log.error("[%s] giving up on restart after 30 attempts",
+                      self.device_name)
 
     def _cleanup_socket(self):
         try:

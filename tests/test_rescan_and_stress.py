@@ -219,3 +219,42 @@ def test_vf_plugin_with_real_smi_binding_degrades(synthetic_host):
         stop.set()
         plugin.stop()
         kubelet.stop()
+
+
+def test_restart_retries_until_kubelet_returns(synthetic_host):
+    """kubelet removes the plugin socket before its Registration
+    service is back: the restart path must retry instead of dying
+    (improvement over the reference, which gives up after one try)."""
+    import os
+    import time
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X",
+        build_kubelet_devices(reg.device_map["75a3"]), reg, config=cfg)
+    plugin.config.connect_timeout_s = 1.0
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        kubelet.wait_register(5)
+        # kubelet "restarts": its socket disappears AND the plugin's
+        # socket is removed
+        kubelet.stop()  # grpc removes its own unix socket file
+        os.remove(plugin.socket_path)
+        time.sleep(2.5)  # let a first re-register attempt fail
+        new_kubelet = StubKubelet(cfg.kubelet_socket)
+        try:
+            req = new_kubelet.wait_register(20)
+            assert req.resource_name == "amd.com/INSTINCT_MI355X"
+        finally:
+            stop.set()
+            plugin.stop()
+            new_kubelet.stop()
+    except BaseException:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
+        raise
