@@ -28,10 +28,12 @@ def main():
     kfd = spec.pop("kfd_nodes_dir")
     cfg = PluginConfig(**spec)
     stop = threading.Event()
+    rescan = threading.Event()
     signal.signal(signal.SIGTERM, lambda *a: stop.set())
     signal.signal(signal.SIGINT, lambda *a: stop.set())
-    initiate_device_plugin(stop_event=stop, config=cfg,
-                           kfd_nodes_dir=kfd,
+    signal.signal(signal.SIGHUP, lambda *a: rescan.set())
+    initiate_device_plugin(stop_event=stop, rescan_event=rescan,
+                           config=cfg, kfd_nodes_dir=kfd,
                            vf_event_watcher_factory=lambda: None)
 
 

@@ -1,0 +1,106 @@
+#!/usr/bin/env python3
+"""Daemon soak: run the plugin for N seconds under continuous Allocate
+traffic, health flips, and periodic SIGHUP rescans; track RSS/fd growth.
+
+Usage: python tools/daemon_soak.py [--seconds 120]
+Exit 0 = no errors and bounded resource growth.
+"""
+
+import argparse
+import os
+import random
+import signal
+import sys
+import tempfile
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from bench_harness.rig import PluginProcess, build_node  # noqa: E402
+from kubevirt_gpu_device_plugin_amd import dpapi  # noqa: E402
+from tests.fixtures import StubKubelet, dial_plugin  # noqa: E402
+
+
+def proc_stats(pid):
+    with open("/proc/%d/status" % pid) as f:
+        rss = next(int(l.split()[1]) for l in f if l.startswith("VmRSS"))
+    fds = len(os.listdir("/proc/%d/fd" % pid))
+    return rss, fds
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seconds", type=float, default=120.0)
+    args = ap.parse_args()
+
+    with tempfile.TemporaryDirectory() as tmp:
+        host = build_node(tmp, 8)
+        cfg = host.config()
+        kubelet = StubKubelet(cfg.kubelet_socket)
+        plugin = PluginProcess(host)
+        rc = 1
+        try:
+            req = kubelet.wait_register(30)
+            sock = os.path.join(cfg.device_plugin_dir, req.endpoint)
+            ch, stub = dial_plugin(sock, timeout=10)
+            devices = [d.ID for d in next(
+                stub.ListAndWatch(dpapi.Empty())).devices]
+            assert len(devices) == 8
+            pid = plugin.proc.pid
+            time.sleep(1.0)
+            rss0, fds0 = proc_stats(pid)
+            allocs = errors = rescans = flips = 0
+            deadline = time.time() + args.seconds
+            last_hup = time.time()
+            rng = random.Random(42)
+            while time.time() < deadline:
+                try:
+                    bdf = rng.choice(devices)
+                    stub.Allocate(dpapi.AllocateRequest(
+                        container_requests=[
+                            dpapi.ContainerAllocateRequest(
+                                devicesIDs=[bdf])]), timeout=5)
+                    allocs += 1
+                except Exception as e:
+                    # rescan restarts the server; transient UNAVAILABLE
+                    # is expected around a HUP
+                    errors += 1
+                    ch.close()
+                    for _ in range(50):
+                        try:
+                            ch, stub = dial_plugin(sock, timeout=5)
+                            break
+                        except Exception:
+                            time.sleep(0.2)
+                if allocs % 20 == 0:
+                    g = rng.choice([str(100 + i) for i in range(8)])
+                    host.remove_vfio_node(g)
+                    host.add_vfio_node(g)
+                    flips += 1
+                if time.time() - last_hup > 10:
+                    plugin.proc.send_signal(signal.SIGHUP)
+                    rescans += 1
+                    last_hup = time.time()
+                    # server restarts; wait for re-registration
+                    kubelet.wait_register(20)
+                    ch.close()
+                    ch, stub = dial_plugin(sock, timeout=10)
+            rss1, fds1 = proc_stats(pid)
+            ch.close()
+            print("soak %.0fs: %d allocs, %d transient errors, "
+                  "%d rescans, %d health flips" %
+                  (args.seconds, allocs, errors, rescans, flips))
+            print("daemon RSS %d→%d kB (Δ%+d), fds %d→%d" %
+                  (rss0, rss1, rss1 - rss0, fds0, fds1))
+            grew = rss1 - rss0 > 20_000 or fds1 - fds0 > 10
+            rc = 1 if grew else 0
+            print("RESULT:", "FAIL (resource growth)" if grew else "OK")
+        finally:
+            plugin.stop()
+            kubelet.stop()
+            host.cleanup()
+        return rc
+
+
+if __name__ == "__main__":
+    sys.exit(main())
