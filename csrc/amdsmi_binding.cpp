@@ -98,6 +98,15 @@ bool available() {
   return try_dlopen() != nullptr;
 }
 
+void smi_shutdown_locked() {
+  if (!g_inited) return;
+  using Fn = amdsmi_status_t (*)(void);
+  sym<Fn>("amdsmi_shut_down")();
+  g_inited = false;
+  g_processors.clear();
+  g_handle_to_index.clear();
+}
+
 void smi_init() {
   std::lock_guard<std::mutex> lk(g_mutex);
   if (g_inited) return;
@@ -112,9 +121,22 @@ void smi_init() {
   auto get_procs = sym<ProcFn>("amdsmi_get_processor_handles");
 
   uint32_t nsock = 0;
-  check(get_sockets(&nsock, nullptr), "amdsmi_get_socket_handles");
+  // A failure mid-enumeration must not leave a half-initialized
+  // library behind — undo the init before rethrowing.
+  try {
+    check(get_sockets(&nsock, nullptr), "amdsmi_get_socket_handles");
+  } catch (...) {
+    smi_shutdown_locked();
+    throw;
+  }
   std::vector<amdsmi_socket_handle> sockets(nsock);
-  check(get_sockets(&nsock, sockets.data()), "amdsmi_get_socket_handles");
+  try {
+    check(get_sockets(&nsock, sockets.data()),
+          "amdsmi_get_socket_handles");
+  } catch (...) {
+    smi_shutdown_locked();
+    throw;
+  }
 
   g_processors.clear();
   g_handle_to_index.clear();
@@ -134,12 +156,7 @@ void smi_init() {
 
 void smi_shutdown() {
   std::lock_guard<std::mutex> lk(g_mutex);
-  if (!g_inited) return;
-  using Fn = amdsmi_status_t (*)(void);
-  sym<Fn>("amdsmi_shut_down")();
-  g_inited = false;
-  g_processors.clear();
-  g_handle_to_index.clear();
+  smi_shutdown_locked();
 }
 
 amdsmi_processor_handle handle_of(uint32_t index) {
