@@ -87,10 +87,15 @@ def preferred_allocation(available_ids, must_include_ids, size,
     def satisfiable(groups, order, key):
         """First group (prioritising groups already holding selected
         devices) whose selected+free total reaches ``size``
-        (reference: generic_device_plugin.go:548-584)."""
+        (reference: generic_device_plugin.go:548-584).  The unknown
+        group (-1) is never packable — the reference likewise treats
+        "no NUMA node found" as falling through to kubelet order
+        (targetNode stays -1, generic_device_plugin.go:549,576)."""
         selected_first = sorted(
             order, key=lambda k: 0 if selected_count(key, k) else 1)
         for k in selected_first:
+            if k == -1:
+                continue
             free = sum(1 for d in groups[k] if d not in chosen)
             if selected_count(key, k) + free >= size:
                 return k
@@ -100,7 +105,7 @@ def preferred_allocation(available_ids, must_include_ids, size,
         # Step 2: one xGMI island, NUMA-packed inside.
         islands, island_order = group_by(island_of)
         k = satisfiable(islands, island_order, island_of)
-        if k is not None and k != -1:
+        if k is not None:
             try_fill_from(islands[k], inner_key=numa_of)
 
     if len(preferred) < size:
