@@ -124,6 +124,9 @@ def main():
                 "vfs_per_gpu": args.vfs_per_gpu,
                 "iommufd": bool(args.iommufd),
                 "p99_us": round(p99_us, 1),
+                "startup_to_allocatable_ms": round(
+                    getattr(measure_allocate, "last_startup_s", 0)
+                    * 1e3, 1),
                 "vf_config_allocatable": vf_counts,
                 "parallelism": "1 plugin process per node",
                 "global_batch": None,

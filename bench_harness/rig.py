@@ -86,6 +86,7 @@ def measure_allocate(n_gpus, steps, warmup, iommufd=False,
                           iommufd=iommufd)
         cfg = host.config()
         kubelet = StubKubelet(cfg.kubelet_socket)
+        t_boot = time.perf_counter()
         plugin = PluginProcess(host)
         try:
             req = kubelet.wait_register(timeout=30.0)
@@ -93,6 +94,10 @@ def measure_allocate(n_gpus, steps, warmup, iommufd=False,
             ch, stub = dial_plugin(sock, timeout=10.0)
             stream = stub.ListAndWatch(dpapi.Empty())
             devices = [d.ID for d in next(stream).devices]
+            # "time to allocatable": daemon exec → discovery → server up
+            # → registered → first full device list streamed
+            measure_allocate.last_startup_s = \
+                time.perf_counter() - t_boot
             expected = n_gpus * vfs_per_gpu if vfs_per_gpu else n_gpus
             assert len(devices) == expected, \
                 "advertised %d != %d" % (len(devices), expected)
