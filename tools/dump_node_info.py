@@ -50,10 +50,35 @@ def smi_dump():
     return out
 
 
+def partition_modes():
+    """MI300-family compute/memory partitioning (CPX/SPX, NPS1/NPS4)
+    per host-driver-resident GPU — affects how many KFD nodes one PF
+    exposes, so it belongs in any topology diagnosis."""
+    import glob
+    out = {}
+    for card in sorted(glob.glob("/sys/class/drm/card*/device")):
+        mode = {}
+        for key in ("current_compute_partition",
+                    "current_memory_partition"):
+            try:
+                with open(os.path.join(card, key)) as f:
+                    mode[key] = f.read().strip()
+            except OSError:
+                pass
+        if mode:
+            try:
+                bdf = os.path.basename(os.readlink(card))
+            except OSError:
+                bdf = card
+            out[bdf] = mode
+    return out
+
+
 def main():
     print(json.dumps({
         "pci_vendor_1002": pci_walk(),
         "kfd_islands": island_map_from_kfd(),
+        "partition_modes": partition_modes(),
         "amdsmi": smi_dump(),
     }, indent=2, default=str))
 
