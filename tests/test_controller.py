@@ -92,3 +92,25 @@ def test_start_failure_tolerated(synthetic_host):
         stop.set()
         ctrl.stop()
         kubelet.stop()
+
+
+def test_mixed_device_ids_two_resource_types(synthetic_host):
+    """MI300X and MI355X on one node → two distinct resource names from
+    pci.ids (reference: one plugin per device id, device_plugin.go:108)."""
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", device_id="75a3", iommu_group="100")
+    h.add_gpu("0000:11:00.0", device_id="74a1", iommu_group="101")
+    ctrl = Controller(config=h.config(), kfd_nodes_dir=h.kfd_nodes)
+    plugins = ctrl.create_plugins()
+    assert sorted(p.device_name for p in plugins) == [
+        "AQUA_VANJARAM_INSTINCT_MI300X", "INSTINCT_MI355X"]
+    assert all(len(p._devs) == 1 for p in plugins)
+
+
+def test_unknown_device_id_uses_raw_hex(synthetic_host):
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", device_id="beef", iommu_group="100")
+    ctrl = Controller(config=h.config(), kfd_nodes_dir=h.kfd_nodes)
+    plugins = ctrl.create_plugins()
+    assert [p.device_name for p in plugins] == ["beef"]
+    assert plugins[0].resource_name == "amd.com/beef"
