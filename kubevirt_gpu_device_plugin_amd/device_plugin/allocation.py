@@ -20,14 +20,24 @@ step 2 degenerates into step 3, i.e. exact reference behavior.
 
 
 def preferred_allocation(available_ids, must_include_ids, size,
-                         numa_of, island_of=None):
+                         numa_of, island_of=None, group_size_of=None):
     """Return the preferred device list (len == ``size`` when possible).
 
     ``numa_of``/``island_of``: callables id -> int (-1 when unknown).
+    ``group_size_of``: callable id -> member count of the device's
+    IOMMU group.  When given, devices in singleton groups are preferred
+    (stable) over co-grouped ones: allocating one member of a shared
+    group binds every sibling's vfio group into the VM, so siblings
+    handed to another pod later would be broken — the reference leaves
+    this hazard entirely to the operator (it only suppresses siblings
+    from env, generic_device_plugin.go:414-420).
     Raises ``ValueError`` when must-include exceeds ``size``.
     """
     if island_of is None:
         island_of = lambda _id: -1  # noqa: E731
+    if group_size_of is not None:
+        # stable: kubelet order is preserved within each group-size tier
+        available_ids = sorted(available_ids, key=group_size_of)
 
     preferred = []
     chosen = set()

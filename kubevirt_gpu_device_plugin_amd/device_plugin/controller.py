@@ -61,6 +61,7 @@ class Controller:
         """Run discovery and instantiate (but not start) all plugin
         servers.  Returns the plugin list."""
         self.registry = discovery.discover(base_path=self.config.pci_base)
+        discovery.warn_shared_groups(self.registry)
         island_of = build_island_lookup(
             self.registry, nodes_dir=self.kfd_nodes_dir)
         log.info("iommu map: %s",

@@ -158,6 +158,20 @@ def discover(base_path=consts.PCI_DEVICES_PATH,
     return reg
 
 
+def warn_shared_groups(reg):
+    """Surface IOMMU groups holding more than one allocatable function:
+    vfio passes whole groups, so allocating one member binds the others
+    into the same VM (island IOMMU grouping must be
+    honored).  GetPreferredAllocation deprioritises these
+    (allocation.py group_size_of)."""
+    for group, devs in reg.iommu_map.items():
+        if len(devs) > 1:
+            log.warning(
+                "IOMMU group %s holds %d allocatable functions (%s): "
+                "they can only be passed through to the SAME VM", group,
+                len(devs), ", ".join(d.addr for d in devs))
+
+
 def _register(reg, dev):
     reg.iommu_map.setdefault(dev.iommu_group, []).append(dev)
     reg.bdf_to_iommu[dev.addr] = dev.iommu_group

@@ -214,6 +214,12 @@ class GenericDevicePlugin(DevicePluginBase):
             numa = {d.ID: (d.topology.nodes[0].ID if d.topology.nodes
                            else -1)
                     for d in self._devs}
+        bdf_to_iommu = self.registry.bdf_to_iommu
+        iommu_map = self.registry.iommu_map
+
+        def group_size_of(bdf):
+            return len(iommu_map.get(bdf_to_iommu.get(bdf), ()))
+
         response = dpapi.PreferredAllocationResponse()
         for req in request.container_requests:
             try:
@@ -222,7 +228,8 @@ class GenericDevicePlugin(DevicePluginBase):
                     list(req.must_include_deviceIDs),
                     int(req.allocation_size),
                     numa_of=lambda i: numa.get(i, -1),
-                    island_of=self.island_of)
+                    island_of=self.island_of,
+                    group_size_of=group_size_of)
             except ValueError as e:
                 context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
             response.container_responses.add(deviceIDs=ids)

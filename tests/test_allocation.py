@@ -88,3 +88,21 @@ def test_full_node_8gpu():
     got = preferred_allocation(avail, [], 8, numa_of=numa,
                                island_of=island)
     assert sorted(got) == avail
+
+
+def test_singleton_groups_preferred_over_shared():
+    """Devices in exclusive IOMMU groups win over co-grouped ones
+    (allocating one member of a shared group drags its siblings)."""
+    numa = numa_map({"a": 0, "b": 0, "c": 0})
+    gsize = {"a": 2, "b": 1, "c": 2}.__getitem__
+    got = preferred_allocation(["a", "b", "c"], [], 1, numa_of=numa,
+                               group_size_of=gsize)
+    assert got == ["b"]
+
+
+def test_group_size_tiebreak_is_stable():
+    numa = numa_map({"a": 0, "b": 0, "c": 0})
+    gsize = {"a": 1, "b": 1, "c": 1}.__getitem__
+    got = preferred_allocation(["a", "b", "c"], [], 2, numa_of=numa,
+                               group_size_of=gsize)
+    assert got == ["a", "b"]  # kubelet order preserved

@@ -326,3 +326,21 @@ def test_pre_start_container(rig, synthetic_host):
         devicesIDs=["0000:0c:00.0"]))
     assert resp == dpapi.PreStartContainerResponse()
     ch.close()
+
+
+def test_preferred_allocation_avoids_shared_group(rig, synthetic_host):
+    """Two co-grouped functions + one exclusive: a 1-device request
+    prefers the exclusive one even though kubelet listed it last."""
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    h.add_gpu("0000:0c:00.1", iommu_group="40")
+    h.add_gpu("0000:2f:00.0", iommu_group="41")
+    _, plugin, _ = rig()
+    ch, stub = dial_plugin(plugin.socket_path)
+    resp = stub.GetPreferredAllocation(dpapi.PreferredAllocationRequest(
+        container_requests=[dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=["0000:0c:00.0", "0000:0c:00.1",
+                                 "0000:2f:00.0"],
+            allocation_size=1)]))
+    assert list(resp.container_responses[0].deviceIDs) == ["0000:2f:00.0"]
+    ch.close()
