@@ -60,8 +60,16 @@ class VfDevicePlugin(GenericDevicePlugin):
         pf_bdfs = sorted({d.parent_pf
                           for devs in self.registry.vf_map.values()
                           for d in devs if d.parent_pf})
+        # Bind THIS server generation's lifetime events now: after a
+        # kubelet-restart cycle self._term is a fresh Event, and a
+        # watcher polling through the swap would otherwise never see
+        # the old one get set and leak a thread per restart.
+        term, stop = self._term, self._stop
         self._event_thread = threading.Thread(
-            target=self._event_loop, args=(watcher, pf_bdfs),
+            target=self._event_loop,
+            args=(watcher, pf_bdfs,
+                  lambda: term.is_set()
+                  or (stop is not None and stop.is_set())),
             name="smi-events-%s" % self.device_name, daemon=True)
         self._event_thread.start()
 
@@ -92,7 +100,7 @@ class VfDevicePlugin(GenericDevicePlugin):
         return [vf for vf in self.registry.pf_vf_map.get(pf_bdf, [])
                 if vf in mine]
 
-    def _event_loop(self, watcher, pf_bdfs):
+    def _event_loop(self, watcher, pf_bdfs, should_stop):
         try:
             watcher.watch(
                 pf_bdfs,
@@ -100,7 +108,7 @@ class VfDevicePlugin(GenericDevicePlugin):
                     self._vfs_of_pf(pf), dpapi.UNHEALTHY),
                 on_healthy=lambda pf: self.set_health(
                     self._vfs_of_pf(pf), dpapi.HEALTHY),
-                should_stop=self._should_exit)
+                should_stop=should_stop)
         except Exception:
             log.exception("[%s] AMD-SMI event loop failed; continuing "
                           "with sysfs-only health", self.device_name)

@@ -139,3 +139,20 @@ def test_vf_vfio_node_health(vf_rig):
     assert health["0000:0c:02.2"] == "Unhealthy"
     assert health["0000:0c:02.0"] == "Healthy"
     ch.close()
+
+
+def test_smi_watcher_stops_across_restart(vf_rig):
+    """A kubelet-restart cycle swaps the server's term event; the old
+    SMI watcher thread must still observe its own generation's stop and
+    exit (no thread leak per restart)."""
+    import threading as _threading
+    smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u0"}])
+    h, pf, plugin, kubelet = vf_rig(n_vfs=2, smi=smi)
+    eventually(lambda: smi.event_inited == {0})
+    plugin.restart()
+    eventually(lambda: smi.event_inited == {0})  # new watcher armed
+
+    def smi_threads():
+        return [t for t in _threading.enumerate()
+                if t.name.startswith("smi-events-") and t.is_alive()]
+    eventually(lambda: len(smi_threads()) == 1, timeout=8.0)
