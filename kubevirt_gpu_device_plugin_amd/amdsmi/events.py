@@ -18,6 +18,7 @@ The 5000 ms poll matches the reference's ``WaitForEvent(…, 5000)``
 """
 
 import logging
+import threading
 
 from . import (EVT_GPU_POST_RESET, EVT_GPU_PRE_RESET,
                EVT_THERMAL_THROTTLE, EVT_VMFAULT, event_mask, is_available,
@@ -109,8 +110,108 @@ class AmdSmiEventWatcher:
             smi.shutdown()
 
 
-def default_watcher():
-    """Factory used by VfDevicePlugin; None ⇒ AMD-SMI unavailable."""
-    if not is_available():
-        return None
-    return AmdSmiEventWatcher()
+class SharedSmiWatcher:
+    """Process-wide multiplexer over ONE AmdSmiEventWatcher thread.
+
+    ``amdsmi_get_gpu_event_notification`` drains a *global* event queue
+    (amdsmi.h:6063-6104): two independent watcher loops would steal
+    each other's events and race init/shut_down.  Plugins therefore
+    subscribe here; one thread watches the union of their PFs and
+    dispatches by BDF.  The thread restarts (generation bump) when the
+    subscriber set changes and exits when it empties.
+    """
+
+    def __init__(self, smi=None, poll_ms=POLL_MS):
+        self._smi = smi
+        self._poll_ms = poll_ms
+        self._lock = threading.Lock()    # protects _subs/_generation
+        self._mgmt = threading.RLock()   # serializes (un)subscribe
+        self._subs = {}       # handle -> (frozenset(pfs), on_un, on_ok)
+        self._next_handle = 0
+        self._generation = 0
+        self._thread = None
+
+    class Subscription:
+        def __init__(self, owner, handle):
+            self._owner = owner
+            self._handle = handle
+
+        def unsubscribe(self):
+            self._owner._unsubscribe(self._handle)
+
+    def subscribe(self, pf_bdfs, on_unhealthy, on_healthy):
+        """Returns a Subscription, or None when AMD-SMI is unavailable
+        on this host (callers degrade to sysfs-only health)."""
+        if self._smi is None and not is_available():
+            return None
+        with self._mgmt:
+            with self._lock:
+                handle = self._next_handle
+                self._next_handle += 1
+                self._subs[handle] = (
+                    frozenset(b.lower() for b in pf_bdfs),
+                    on_unhealthy, on_healthy)
+            self._restart()
+        return self.Subscription(self, handle)
+
+    def _unsubscribe(self, handle):
+        with self._mgmt:
+            with self._lock:
+                self._subs.pop(handle, None)
+            self._restart()
+
+    def _restart(self):
+        """Called under _mgmt: retire the running generation, join it,
+        start a new one for the current subscriber union."""
+        with self._lock:
+            self._generation += 1
+            gen = self._generation
+            old = self._thread
+            subs = list(self._subs.values())
+        if old is not None and old.is_alive() \
+                and old is not threading.current_thread():
+            # the old generation notices the bump within one poll;
+            # join so two loops never touch the amdsmi event API at once
+            old.join(timeout=30.0)
+        self._thread = None
+        if not subs:
+            return
+        pfs = sorted(set().union(*(s[0] for s in subs)))
+        t = threading.Thread(target=self._run, args=(gen, pfs),
+                             name="smi-events-shared", daemon=True)
+        self._thread = t
+        t.start()
+
+    def _dispatch(self, which, pf):
+        pf = pf.lower()
+        with self._lock:
+            subs = list(self._subs.values())
+        for pfs, on_un, on_ok in subs:
+            if pf in pfs:
+                (on_un if which == "unhealthy" else on_ok)(pf)
+
+    def _run(self, gen, pfs):
+        watcher = AmdSmiEventWatcher(smi=self._smi,
+                                     poll_ms=self._poll_ms)
+        try:
+            watcher.watch(
+                pfs,
+                on_unhealthy=lambda pf: self._dispatch("unhealthy", pf),
+                on_healthy=lambda pf: self._dispatch("healthy", pf),
+                should_stop=lambda: gen != self._generation)
+        except Exception:
+            log.exception("shared AMD-SMI watcher failed; subscribers "
+                          "fall back to sysfs-only health")
+
+
+_shared = None
+_shared_lock = threading.Lock()
+
+
+def shared_watcher():
+    """The module-level SharedSmiWatcher used by VF plugins."""
+    global _shared
+    with _shared_lock:
+        if _shared is None:
+            _shared = SharedSmiWatcher()
+        return _shared

@@ -26,7 +26,6 @@ sysfs-only health, matching the reference's NVML-absent behavior
 """
 
 import logging
-import threading
 
 from .. import dpapi
 from . import consts
@@ -37,41 +36,51 @@ log = logging.getLogger(__name__)
 
 class VfDevicePlugin(GenericDevicePlugin):
     def __init__(self, device_name, devices, registry, config=None,
-                 island_of=None, event_watcher_factory=None):
+                 island_of=None, smi_watcher=None,
+                 event_watcher_factory=None):
         super().__init__(device_name, devices, registry, config=config,
                          island_of=island_of,
                          env_prefix=consts.VF_ENV_PREFIX)
-        # factory() -> object with watch(pf_bdfs, on_unhealthy,
-        # on_healthy, should_stop); injectable for tests
-        # (reference seam: watchXIDs var, generic_vgpu_device_plugin.go:47).
+        # smi_watcher: a SharedSmiWatcher (None → process-wide default).
+        # All VF plugins share ONE watcher thread because the amdsmi
+        # event queue is global — independent pollers would steal each
+        # other's events (see amdsmi/events.py SharedSmiWatcher).
+        # event_watcher_factory=lambda: None disables SMI health
+        # entirely (test/bench seam; reference analogue: watchXIDs var,
+        # generic_vgpu_device_plugin.go:47).
+        self._smi_watcher = smi_watcher
         self._event_watcher_factory = event_watcher_factory
-        self._event_thread = None
+        self._smi_sub = None
 
     def start(self, stop_event):
         super().start(stop_event)
-        if self._event_watcher_factory is None:
-            from ..amdsmi import events as smi_events
-            self._event_watcher_factory = smi_events.default_watcher
-        watcher = self._event_watcher_factory()
-        if watcher is None:
-            log.warning("[%s] AMD-SMI unavailable; VF health relies on "
-                        "vfio node watching only", self.device_name)
+        if self._event_watcher_factory is not None \
+                and self._event_watcher_factory() is None:
+            log.info("[%s] SMI health disabled by configuration",
+                     self.device_name)
             return
+        watcher = self._smi_watcher
+        if watcher is None:
+            from ..amdsmi import events as smi_events
+            watcher = smi_events.shared_watcher()
         pf_bdfs = sorted({d.parent_pf
                           for devs in self.registry.vf_map.values()
                           for d in devs if d.parent_pf})
-        # Bind THIS server generation's lifetime events now: after a
-        # kubelet-restart cycle self._term is a fresh Event, and a
-        # watcher polling through the swap would otherwise never see
-        # the old one get set and leak a thread per restart.
-        term, stop = self._term, self._stop
-        self._event_thread = threading.Thread(
-            target=self._event_loop,
-            args=(watcher, pf_bdfs,
-                  lambda: term.is_set()
-                  or (stop is not None and stop.is_set())),
-            name="smi-events-%s" % self.device_name, daemon=True)
-        self._event_thread.start()
+        self._smi_sub = watcher.subscribe(
+            pf_bdfs,
+            on_unhealthy=lambda pf: self.set_health(
+                self._vfs_of_pf(pf), dpapi.UNHEALTHY),
+            on_healthy=lambda pf: self.set_health(
+                self._vfs_of_pf(pf), dpapi.HEALTHY))
+        if self._smi_sub is None:
+            log.warning("[%s] AMD-SMI unavailable; VF health relies on "
+                        "vfio node watching only", self.device_name)
+
+    def stop(self):
+        sub, self._smi_sub = self._smi_sub, None
+        if sub is not None:
+            sub.unsubscribe()
+        super().stop()
 
     def Allocate(self, request, context):  # noqa: N802
         """VF allocation = passthrough allocation, plus resource-type
@@ -100,15 +109,3 @@ class VfDevicePlugin(GenericDevicePlugin):
         return [vf for vf in self.registry.pf_vf_map.get(pf_bdf, [])
                 if vf in mine]
 
-    def _event_loop(self, watcher, pf_bdfs, should_stop):
-        try:
-            watcher.watch(
-                pf_bdfs,
-                on_unhealthy=lambda pf: self.set_health(
-                    self._vfs_of_pf(pf), dpapi.UNHEALTHY),
-                on_healthy=lambda pf: self.set_health(
-                    self._vfs_of_pf(pf), dpapi.HEALTHY),
-                should_stop=should_stop)
-        except Exception:
-            log.exception("[%s] AMD-SMI event loop failed; continuing "
-                          "with sysfs-only health", self.device_name)

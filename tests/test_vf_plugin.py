@@ -9,7 +9,9 @@ from kubevirt_gpu_device_plugin_amd import dpapi
 from kubevirt_gpu_device_plugin_amd.amdsmi import (
     EVT_GPU_POST_RESET, EVT_GPU_PRE_RESET, EVT_VMFAULT,
 )
-from kubevirt_gpu_device_plugin_amd.amdsmi.events import AmdSmiEventWatcher
+from kubevirt_gpu_device_plugin_amd.amdsmi.events import (
+    AmdSmiEventWatcher, SharedSmiWatcher,
+)
 from kubevirt_gpu_device_plugin_amd.device_plugin import discovery
 from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
     build_kubelet_devices,
@@ -39,12 +41,14 @@ def vf_rig(synthetic_host):
         cfg = h.config()
         kubelet = StubKubelet(cfg.kubelet_socket)
         reg = discovery.discover(base_path=h.pci)
-        factory = (lambda: AmdSmiEventWatcher(smi=smi, poll_ms=50)) \
-            if smi is not None else (lambda: None)
+        if smi is not None:
+            kw = dict(smi_watcher=SharedSmiWatcher(smi=smi, poll_ms=50))
+        else:
+            kw = dict(event_watcher_factory=lambda: None)
         plugin = VfDevicePlugin(
             "INSTINCT_MI355X_VF",
             build_kubelet_devices(reg.vf_map["75b3"]), reg, config=cfg,
-            event_watcher_factory=factory)
+            **kw)
         stop = threading.Event()
         plugin.start(stop)
         created.update(plugin=plugin, kubelet=kubelet, stop=stop)
@@ -141,18 +145,62 @@ def test_vf_vfio_node_health(vf_rig):
     ch.close()
 
 
-def test_smi_watcher_stops_across_restart(vf_rig):
-    """A kubelet-restart cycle swaps the server's term event; the old
-    SMI watcher thread must still observe its own generation's stop and
-    exit (no thread leak per restart)."""
+def test_smi_watcher_survives_restart_one_thread(vf_rig):
+    """Across a kubelet-restart cycle there is exactly ONE shared SMI
+    watcher thread, and events still reach the restarted server."""
     import threading as _threading
     smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u0"}])
     h, pf, plugin, kubelet = vf_rig(n_vfs=2, smi=smi)
     eventually(lambda: smi.event_inited == {0})
     plugin.restart()
-    eventually(lambda: smi.event_inited == {0})  # new watcher armed
+    eventually(lambda: smi.event_inited == {0})  # re-armed
 
     def smi_threads():
         return [t for t in _threading.enumerate()
-                if t.name.startswith("smi-events-") and t.is_alive()]
+                if t.name == "smi-events-shared" and t.is_alive()]
     eventually(lambda: len(smi_threads()) == 1, timeout=8.0)
+
+    smi.push(0, EVT_GPU_PRE_RESET, "after restart")
+    eventually(lambda: all(
+        d.health == dpapi.UNHEALTHY for d in plugin.devices_snapshot()))
+
+
+def test_two_vf_plugins_share_one_watcher(synthetic_host):
+    """Two VF resource types on one node: a single shared watcher
+    dispatches a PF fault only to the plugin owning that PF."""
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", driver="gim", iommu_group="40")
+    h.add_vf("0000:0c:02.0", pf_bdf="0000:0c:00.0", iommu_group="50")
+    h.add_gpu("0000:0d:00.0", driver="gim", iommu_group="41")
+    h.add_vf("0000:0d:02.0", pf_bdf="0000:0d:00.0", device_id="75b0",
+             iommu_group="51")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "a"},
+                   {"index": 1, "bdf": "0000:0d:00.0", "uuid": "b"}])
+    shared = SharedSmiWatcher(smi=smi, poll_ms=50)
+    stop = threading.Event()
+    plugins = []
+    try:
+        for name, dev_id in (("INSTINCT_MI355X_VF", "75b3"),
+                             ("INSTINCT_MI350X_VF", "75b0")):
+            p = VfDevicePlugin(name,
+                               build_kubelet_devices(reg.vf_map[dev_id]),
+                               reg, config=cfg, smi_watcher=shared)
+            p.start(stop)
+            plugins.append(p)
+        eventually(lambda: smi.event_inited == {0, 1})
+        import threading as _threading
+        assert len([t for t in _threading.enumerate()
+                    if t.name == "smi-events-shared"
+                    and t.is_alive()]) == 1
+        smi.push(1, EVT_GPU_PRE_RESET, "pf 0d down")
+        eventually(lambda: plugins[1].devices_snapshot()[0].health
+                   == dpapi.UNHEALTHY)
+        assert plugins[0].devices_snapshot()[0].health == dpapi.HEALTHY
+    finally:
+        stop.set()
+        for p in plugins:
+            p.stop()
+        kubelet.stop()
