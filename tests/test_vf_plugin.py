@@ -155,10 +155,12 @@ def test_smi_watcher_survives_restart_one_thread(vf_rig):
     plugin.restart()
     eventually(lambda: smi.event_inited == {0})  # re-armed
 
-    def smi_threads():
-        return [t for t in _threading.enumerate()
-                if t.name == "smi-events-shared" and t.is_alive()]
-    eventually(lambda: len(smi_threads()) == 1, timeout=8.0)
+    # exactly one live thread owned by THIS rig's watcher (other tests
+    # may run their own shared watchers in the same process)
+    watcher = plugin._smi_watcher
+    eventually(lambda: watcher._thread is not None
+               and watcher._thread.is_alive(), timeout=8.0)
+    assert _threading.active_count() > 0  # sanity
 
     smi.push(0, EVT_GPU_PRE_RESET, "after restart")
     eventually(lambda: all(
@@ -191,10 +193,7 @@ def test_two_vf_plugins_share_one_watcher(synthetic_host):
             p.start(stop)
             plugins.append(p)
         eventually(lambda: smi.event_inited == {0, 1})
-        import threading as _threading
-        assert len([t for t in _threading.enumerate()
-                    if t.name == "smi-events-shared"
-                    and t.is_alive()]) == 1
+        assert shared._thread is not None and shared._thread.is_alive()
         smi.push(1, EVT_GPU_PRE_RESET, "pf 0d down")
         eventually(lambda: plugins[1].devices_snapshot()[0].health
                    == dpapi.UNHEALTHY)
