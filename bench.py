@@ -39,6 +39,9 @@ def main():
     ap.add_argument("--vfs-per-gpu", type=int, default=0,
                     help="SR-IOV mode: N gim PFs x this many VFs; each "
                          "RPC allocates one PF's VF set")
+    ap.add_argument("--single", action="store_true",
+                    help="allocate 1 device per RPC round-robin "
+                         "(pod-per-GPU pattern) instead of all N")
     ap.add_argument("--vf-check", action="store_true",
                     help="also verify the 64-VF SR-IOV config counts")
     args = ap.parse_args()
@@ -75,7 +78,8 @@ def main():
         from bench_harness.rig import measure_allocate
         lat, n_devices = measure_allocate(
             args.gpus, args.steps, args.warmup, iommufd=args.iommufd,
-            vfs_per_gpu=args.vfs_per_gpu)
+            vfs_per_gpu=args.vfs_per_gpu,
+            allocate_all=not args.single)
     elapsed = time.perf_counter() - t_start
     barrier_sync()
 
@@ -119,7 +123,8 @@ def main():
                 "allocatable_gpus": n_devices,
                 "allocate_request": (
                     "%d VFs (one PF set) per RPC" % args.vfs_per_gpu
-                    if args.vfs_per_gpu
+                    if args.vfs_per_gpu else
+                    "1 device per RPC (round-robin)" if args.single
                     else "all %d devices per RPC" % n_devices),
                 "vfs_per_gpu": args.vfs_per_gpu,
                 "iommufd": bool(args.iommufd),
