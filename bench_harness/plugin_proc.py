@@ -33,6 +33,7 @@ def main():
     signal.signal(signal.SIGINT, lambda *a: stop.set())
     signal.signal(signal.SIGHUP, lambda *a: rescan.set())
     initiate_device_plugin(stop_event=stop, rescan_event=rescan,
+                           uevent_autoscan=False,  # deterministic rig
                            config=cfg, kfd_nodes_dir=kfd,
                            vf_event_watcher_factory=lambda: None)
 

@@ -109,25 +109,33 @@ class Controller:
                 log.exception("error stopping %s", p.device_name)
 
 
-def initiate_device_plugin(stop_event=None, rescan_event=None, **kwargs):
+def initiate_device_plugin(stop_event=None, rescan_event=None,
+                           uevent_autoscan=True, **kwargs):
     """Blocking entry point (reference: InitiateDevicePlugin,
     device_plugin.go:89-96).
 
-    ``rescan_event`` (set by SIGHUP in cmd/main.py) triggers a full
-    re-discovery + re-registration cycle.  The reference discovers
-    exactly once and needs a process restart to pick up new devices
-    (SURVEY.md §5 "no hotplug re-scan") — on MI355X that matters:
-    gim instantiates VFs *after* daemon start (``echo 8 >
-    sriov_numvfs``), so an operator can HUP the daemon instead of
-    bouncing it.
+    ``rescan_event`` triggers a full re-discovery + re-registration
+    cycle; it is set by SIGHUP (cmd/main.py) and — when
+    ``uevent_autoscan`` and the netlink socket are available — by
+    kernel PCI uevents for vendor-1002 driver bind/unbind (gim creating
+    VFs, driverctl overrides).  The reference discovers exactly once
+    and needs a process restart to pick up new devices (SURVEY.md §5
+    "no hotplug re-scan").
     """
     stop_event = stop_event or threading.Event()
+    rescan_event = rescan_event if rescan_event is not None \
+        else threading.Event()
+    if uevent_autoscan:
+        from . import uevent
+        uevent.start_listener(rescan_event, stop_event.is_set)
     controller = Controller(**kwargs)
     controller.create_plugins()
     controller.start(stop_event)
     while not stop_event.is_set():
         stop_event.wait(0.5)
-        if rescan_event is not None and rescan_event.is_set():
+        if rescan_event.is_set():
+            # debounce: a gim VF burst emits one uevent per function
+            stop_event.wait(0.5)
             rescan_event.clear()
             log.info("rescan requested: re-running discovery")
             controller.stop()
