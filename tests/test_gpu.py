@@ -140,3 +140,39 @@ def test_real_pci_walk_finds_amd_devices():
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+def test_uevent_autorescan_end_to_end():
+    """Synthesize a kernel uevent on the real MI355X (write 'add' to
+    its sysfs uevent file) and observe the listener request a rescan —
+    the full hotplug path minus actual driver rebinding."""
+    import threading
+    import time
+
+    from kubevirt_gpu_device_plugin_amd import _sysfs
+    from kubevirt_gpu_device_plugin_amd.device_plugin import uevent
+
+    recs = _sysfs.scan_pci("/sys/bus/pci/devices", "1002")
+    gpus = [r for r in recs if r["driver"] == "amdgpu"]
+    if not gpus:
+        pytest.skip("no amdgpu-bound AMD device to poke")
+    rescan = threading.Event()
+    stop = threading.Event()
+    try:
+        listener = uevent.UeventListener(rescan)
+    except OSError as e:
+        pytest.skip("netlink unavailable: %s" % e)
+    t = listener.start(stop.is_set)
+    try:
+        time.sleep(0.2)  # listener armed
+        path = "/sys/bus/pci/devices/%s/uevent" % gpus[0]["addr"]
+        try:
+            with open(path, "w") as f:
+                f.write("add\n")
+        except OSError as e:
+            pytest.skip("cannot synthesize uevent: %s" % e)
+        assert rescan.wait(timeout=10.0), \
+            "no rescan triggered by synthesized uevent"
+    finally:
+        stop.set()
+        t.join(timeout=5)
