@@ -27,6 +27,8 @@ sysfs-only health, matching the reference's NVML-absent behavior
 
 import logging
 
+import grpc
+
 from .. import dpapi
 from . import consts
 from .plugin import GenericDevicePlugin
@@ -91,8 +93,6 @@ class VfDevicePlugin(GenericDevicePlugin):
         (generic_vgpu_device_plugin.go:218-223); a loud INVALID_ARGUMENT
         beats a VM that boots without its GPU.
         """
-        import grpc
-
         mine = {d.ID for d in self._devs}
         for req in request.container_requests:
             for bdf in req.devicesIDs:
