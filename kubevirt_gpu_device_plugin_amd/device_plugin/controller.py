@@ -108,6 +108,64 @@ class Controller:
             except Exception:
                 log.exception("error stopping %s", p.device_name)
 
+    def _desired_resources(self):
+        """name -> (kind, discovery devices) for the current registry."""
+        desired = {}
+        for device_id, devs in sorted(self.registry.device_map.items()):
+            desired[resolve_name(device_id, self.pci_ids_path)] = \
+                ("gpu", devs)
+        for device_id, devs in sorted(self.registry.vf_map.items()):
+            desired[resolve_name(device_id, self.pci_ids_path)] = \
+                ("vf", devs)
+        return desired
+
+    def rescan(self, stop_event):
+        """Diff-based re-discovery (hotplug): resource types that still
+        exist get an in-place device-list update over their live
+        ListAndWatch stream — no socket churn, no re-registration;
+        vanished types stop; new types start and register.  The common
+        case (gim VF count change on an existing type) is therefore
+        zero-disruption.  The reference has no rescan at all
+        (SURVEY.md §5)."""
+        self.registry = discovery.discover(base_path=self.config.pci_base)
+        discovery.warn_shared_groups(self.registry)
+        island_of = build_island_lookup(
+            self.registry, nodes_dir=self.kfd_nodes_dir)
+        desired = self._desired_resources()
+        kept = []
+        for p in self.plugins:
+            if p.device_name in desired:
+                _, devs = desired.pop(p.device_name)
+                p.update_registry(self.registry,
+                                  build_kubelet_devices(devs),
+                                  island_of=island_of)
+                kept.append(p)
+            else:
+                log.info("resource %s vanished; stopping its server",
+                         p.device_name)
+                try:
+                    p.stop()
+                except Exception:
+                    log.exception("error stopping %s", p.device_name)
+        for name, (kind, devs) in desired.items():
+            if kind == "gpu":
+                p = GenericDevicePlugin(
+                    name, build_kubelet_devices(devs), self.registry,
+                    config=self.config, island_of=island_of)
+            else:
+                p = VfDevicePlugin(
+                    name, build_kubelet_devices(devs), self.registry,
+                    config=self.config, island_of=island_of,
+                    event_watcher_factory=self.vf_event_watcher_factory)
+            try:
+                p.start(stop_event)
+                kept.append(p)
+            except Exception as e:
+                log.error("error starting %s device plugin: %s",
+                          name, e)
+        self.plugins = kept
+        return kept
+
 
 def initiate_device_plugin(stop_event=None, rescan_event=None,
                            uevent_autoscan=True, **kwargs):
@@ -138,8 +196,6 @@ def initiate_device_plugin(stop_event=None, rescan_event=None,
             stop_event.wait(0.5)
             rescan_event.clear()
             log.info("rescan requested: re-running discovery")
-            controller.stop()
-            controller.create_plugins()
-            controller.start(stop_event)
+            controller.rescan(stop_event)
     log.info("shutting down device plugin controller")
     controller.stop()

@@ -55,6 +55,15 @@ class GenericDevicePlugin(DevicePluginBase):
             self._spec_cache[host_path] = s
         return s
 
+    def update_registry(self, registry, devices, island_of=None):
+        """Swap in a fresh post-rescan registry + device list.  The
+        registry reference swap is atomic (Allocate readers pick up
+        either the old or the new immutable registry)."""
+        self.registry = registry
+        if island_of is not None:
+            self.island_of = island_of
+        self.update_devices(devices)
+
     # ---- health wiring --------------------------------------------------
 
     def _group_to_ids(self):

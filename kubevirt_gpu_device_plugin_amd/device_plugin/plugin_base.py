@@ -210,6 +210,18 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             return [dpapi.Device.FromString(d.SerializeToString())
                     for d in self._devs]
 
+    def update_devices(self, devices):
+        """Replace the advertised device list in place (hotplug rescan).
+        ListAndWatch streams the new full list; no socket churn or
+        re-registration — kubelet handles inventory changes over the
+        existing stream."""
+        with self._lock:
+            self._devs = list(devices)
+            self._version += 1
+            self._lock.notify_all()
+        log.info("[%s] device list updated: %d devices",
+                 self.device_name, len(devices))
+
     def set_health(self, device_ids, health):
         """Flip health for the given advertised IDs and wake ListAndWatch
         streams (takes the role of the healthy/unhealthy channels,
@@ -281,7 +293,6 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         """inotify loop: vfio node create/remove → health flips; removal
         of our own socket → kubelet restarted → full server restart
         (reference: healthCheck, generic_device_plugin.go:619-697)."""
-        group_to_ids = self._group_to_ids()
         sock_base = os.path.basename(self.socket_path)
         with inotify.Watcher() as w:
             w.add_watch(self.config.device_plugin_dir)
@@ -292,7 +303,13 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                             self.config.vfio_dir, e)
             self._watch_armed.set()
             while not self._should_exit():
-                for ev in w.read_events(timeout_s=0.2):
+                events = w.read_events(timeout_s=0.2)
+                if not events:
+                    continue
+                # recomputed per batch: device lists change in place on
+                # hotplug rescans (update_devices)
+                group_to_ids = self._group_to_ids()
+                for ev in events:
                     if self._should_exit():
                         return
                     path = w.path_of(ev.wd)

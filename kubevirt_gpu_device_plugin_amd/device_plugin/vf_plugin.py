@@ -84,6 +84,26 @@ class VfDevicePlugin(GenericDevicePlugin):
             sub.unsubscribe()
         super().stop()
 
+    def update_registry(self, registry, devices, island_of=None):
+        super().update_registry(registry, devices, island_of=island_of)
+        # PF set may have changed (new gim PFs) — resubscribe
+        sub, self._smi_sub = self._smi_sub, None
+        if sub is not None:
+            sub.unsubscribe()
+            pf_bdfs = sorted({d.parent_pf
+                              for devs in registry.vf_map.values()
+                              for d in devs if d.parent_pf})
+            watcher = self._smi_watcher
+            if watcher is None:
+                from ..amdsmi import events as smi_events
+                watcher = smi_events.shared_watcher()
+            self._smi_sub = watcher.subscribe(
+                pf_bdfs,
+                on_unhealthy=lambda pf: self.set_health(
+                    self._vfs_of_pf(pf), dpapi.UNHEALTHY),
+                on_healthy=lambda pf: self.set_health(
+                    self._vfs_of_pf(pf), dpapi.HEALTHY))
+
     def Allocate(self, request, context):  # noqa: N802
         """VF allocation = passthrough allocation, plus resource-type
         validation: every requested BDF must be a VF of *this* type.

@@ -39,10 +39,10 @@ def test_rescan_picks_up_new_vfs(synthetic_host):
         h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0",
                  iommu_group="120")
         rescan.set()
-        names = {kubelet.wait_register(10).resource_name
-                 for _ in range(2)}
-        assert names == {"amd.com/INSTINCT_MI355X",
-                         "amd.com/INSTINCT_MI355X_VF"}
+        # diff-rescan: only the NEW type registers; the existing
+        # passthrough resource keeps its socket and registration
+        req = kubelet.wait_register(10)
+        assert req.resource_name == "amd.com/INSTINCT_MI355X_VF"
     finally:
         stop.set()
         t.join(timeout=10)
@@ -258,3 +258,87 @@ def test_restart_retries_until_kubelet_returns(synthetic_host):
         plugin.stop()
         kubelet.stop()
         raise
+
+
+def test_rescan_updates_existing_type_in_place(synthetic_host):
+    """VF count change on an existing resource type: the live
+    ListAndWatch stream gets the new inventory — same socket, no
+    re-registration."""
+    from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
+        Controller,
+    )
+    h = synthetic_host
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0", iommu_group="120")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    ctrl = Controller(config=cfg, kfd_nodes_dir=h.kfd_nodes,
+                      vf_event_watcher_factory=lambda: None)
+    ctrl.create_plugins()
+    stop = threading.Event()
+    try:
+        started = ctrl.start(stop)
+        kubelet.wait_register(10)
+        plugin = started[0]
+        ch, stub = dial_plugin(plugin.socket_path)
+        stream = stub.ListAndWatch(dpapi.Empty())
+        assert len(next(stream).devices) == 1
+
+        # operator raises sriov_numvfs: 3 more VFs appear
+        for i in (1, 2, 3):
+            h.add_vf("0000:20:02.%d" % i, pf_bdf="0000:20:00.0",
+                     iommu_group=str(120 + i))
+        ctrl.rescan(stop)
+        upd = next(stream)  # SAME stream — no socket churn
+        assert len(upd.devices) == 4
+        # no second registration happened
+        assert kubelet.requests.empty()
+        # new VF is allocatable immediately
+        resp = stub.Allocate(dpapi.AllocateRequest(
+            container_requests=[dpapi.ContainerAllocateRequest(
+                devicesIDs=["0000:20:02.3"])]))
+        assert resp.container_responses[0].envs
+        # health watch covers the new VF too
+        h.remove_vfio_node("123")
+        upd = next(stream)
+        assert {d.ID: d.health for d in upd.devices}[
+            "0000:20:02.3"] == "Unhealthy"
+        ch.close()
+    finally:
+        stop.set()
+        ctrl.stop()
+        kubelet.stop()
+
+
+def test_rescan_adds_and_removes_types(synthetic_host):
+    """A vanished type's server stops; a new type registers."""
+    from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
+        Controller,
+    )
+    import os
+    import shutil
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", iommu_group="100")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    ctrl = Controller(config=cfg, kfd_nodes_dir=h.kfd_nodes,
+                      vf_event_watcher_factory=lambda: None)
+    ctrl.create_plugins()
+    stop = threading.Event()
+    try:
+        started = ctrl.start(stop)
+        kubelet.wait_register(10)
+        gpu_sock = started[0].socket_path
+        # the GPU is unbound; a different-type device appears
+        shutil.rmtree(os.path.join(h.pci, "0000:10:00.0"))
+        h.add_gpu("0000:11:00.0", device_id="74a1", iommu_group="101")
+        ctrl.rescan(stop)
+        req = kubelet.wait_register(10)
+        assert req.resource_name == "amd.com/AQUA_VANJARAM_INSTINCT_MI300X"
+        assert not os.path.exists(gpu_sock)  # vanished type stopped
+        assert [p.device_name for p in ctrl.plugins] == [
+            "AQUA_VANJARAM_INSTINCT_MI300X"]
+    finally:
+        stop.set()
+        ctrl.stop()
+        kubelet.stop()
