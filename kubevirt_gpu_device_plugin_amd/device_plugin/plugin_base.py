@@ -214,9 +214,14 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         """Replace the advertised device list in place (hotplug rescan).
         ListAndWatch streams the new full list; no socket churn or
         re-registration — kubelet handles inventory changes over the
-        existing stream."""
+        existing stream.  Devices that persist keep their current
+        health (a rescan must not quietly resurrect an Unhealthy GPU)."""
         with self._lock:
+            old_health = {d.ID: d.health for d in self._devs}
             self._devs = list(devices)
+            for d in self._devs:
+                if d.ID in old_health:
+                    d.health = old_health[d.ID]
             self._version += 1
             self._lock.notify_all()
         log.info("[%s] device list updated: %d devices",

@@ -342,3 +342,37 @@ def test_rescan_adds_and_removes_types(synthetic_host):
         stop.set()
         ctrl.stop()
         kubelet.stop()
+
+
+def test_rescan_preserves_health_state(synthetic_host):
+    """An Unhealthy device must stay Unhealthy through a rescan."""
+    from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
+        Controller,
+    )
+    h = synthetic_host
+    h.add_gpu("0000:10:00.0", iommu_group="100")
+    h.add_gpu("0000:11:00.0", iommu_group="101")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    ctrl = Controller(config=cfg, kfd_nodes_dir=h.kfd_nodes,
+                      vf_event_watcher_factory=lambda: None)
+    ctrl.create_plugins()
+    stop = threading.Event()
+    try:
+        started = ctrl.start(stop)
+        kubelet.wait_register(10)
+        plugin = started[0]
+        h.remove_vfio_node("100")
+        eventually(lambda: {d.ID: d.health
+                            for d in plugin.devices_snapshot()}[
+            "0000:10:00.0"] == "Unhealthy")
+        h.add_gpu("0000:12:00.0", iommu_group="102")  # hotplug
+        ctrl.rescan(stop)
+        health = {d.ID: d.health for d in plugin.devices_snapshot()}
+        assert health == {"0000:10:00.0": "Unhealthy",
+                          "0000:11:00.0": "Healthy",
+                          "0000:12:00.0": "Healthy"}
+    finally:
+        stop.set()
+        ctrl.stop()
+        kubelet.stop()
