@@ -75,19 +75,21 @@ class Controller:
                   for k, v in self.registry.vf_map.items()})
         log.info("pf→vf map: %s", self.registry.pf_vf_map)
 
-        self.plugins = []
-        for device_id, devs in sorted(self.registry.device_map.items()):
-            name = resolve_name(device_id, self.pci_ids_path)
-            self.plugins.append(GenericDevicePlugin(
-                name, build_kubelet_devices(devs), self.registry,
-                config=self.config, island_of=island_of))
-        for device_id, devs in sorted(self.registry.vf_map.items()):
-            name = resolve_name(device_id, self.pci_ids_path)
-            self.plugins.append(VfDevicePlugin(
-                name, build_kubelet_devices(devs), self.registry,
-                config=self.config, island_of=island_of,
-                event_watcher_factory=self.vf_event_watcher_factory))
+        self.plugins = [
+            self._make_plugin(name, kind, devs, island_of)
+            for name, (kind, devs) in self._desired_resources().items()
+        ]
         return self.plugins
+
+    def _make_plugin(self, name, kind, devs, island_of):
+        if kind == "gpu":
+            return GenericDevicePlugin(
+                name, build_kubelet_devices(devs), self.registry,
+                config=self.config, island_of=island_of)
+        return VfDevicePlugin(
+            name, build_kubelet_devices(devs), self.registry,
+            config=self.config, island_of=island_of,
+            event_watcher_factory=self.vf_event_watcher_factory)
 
     def start(self, stop_event):
         started = []
@@ -148,15 +150,7 @@ class Controller:
                 except Exception:
                     log.exception("error stopping %s", p.device_name)
         for name, (kind, devs) in desired.items():
-            if kind == "gpu":
-                p = GenericDevicePlugin(
-                    name, build_kubelet_devices(devs), self.registry,
-                    config=self.config, island_of=island_of)
-            else:
-                p = VfDevicePlugin(
-                    name, build_kubelet_devices(devs), self.registry,
-                    config=self.config, island_of=island_of,
-                    event_watcher_factory=self.vf_event_watcher_factory)
+            p = self._make_plugin(name, kind, devs, island_of)
             try:
                 p.start(stop_event)
                 kept.append(p)
