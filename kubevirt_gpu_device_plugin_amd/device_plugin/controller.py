@@ -166,8 +166,8 @@ def initiate_device_plugin(stop_event=None, rescan_event=None,
     """Blocking entry point (reference: InitiateDevicePlugin,
     device_plugin.go:89-96).
 
-    ``rescan_event`` triggers a full re-discovery + re-registration
-    cycle; it is set by SIGHUP (cmd/main.py) and — when
+    ``rescan_event`` triggers a diff-based re-discovery (see
+    Controller.rescan); it is set by SIGHUP (cmd/main.py) and — when
     ``uevent_autoscan`` and the netlink socket are available — by
     kernel PCI uevents for vendor-1002 driver bind/unbind (gim creating
     VFs, driverctl overrides).  The reference discovers exactly once

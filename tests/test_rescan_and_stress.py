@@ -376,3 +376,52 @@ def test_rescan_preserves_health_state(synthetic_host):
         stop.set()
         ctrl.stop()
         kubelet.stop()
+
+
+def test_rescan_resubscribes_smi_for_new_pf(synthetic_host):
+    """A new gim PF (with VFs of an existing type) appears at rescan:
+    the SMI subscription must cover it — a fault on the NEW PF flips
+    its VFs."""
+    from kubevirt_gpu_device_plugin_amd.amdsmi import EVT_GPU_PRE_RESET
+    from kubevirt_gpu_device_plugin_amd.amdsmi.events import (
+        SharedSmiWatcher,
+    )
+    from kubevirt_gpu_device_plugin_amd.device_plugin.vf_plugin import (
+        VfDevicePlugin,
+    )
+    from kubevirt_gpu_device_plugin_amd import dpapi as _dpapi
+    from tests.fixtures import FakeSmi
+    h = synthetic_host
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    h.add_vf("0000:20:02.0", pf_bdf="0000:20:00.0", iommu_group="120")
+    cfg = h.config()
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    smi = FakeSmi([{"index": 0, "bdf": "0000:20:00.0", "uuid": "a"},
+                   {"index": 1, "bdf": "0000:21:00.0", "uuid": "b"}])
+    shared = SharedSmiWatcher(smi=smi, poll_ms=50)
+    reg = discovery.discover(base_path=h.pci)
+    plugin = VfDevicePlugin(
+        "INSTINCT_MI355X_VF",
+        build_kubelet_devices(reg.vf_map["75b3"]), reg, config=cfg,
+        smi_watcher=shared)
+    stop = threading.Event()
+    plugin.start(stop)
+    try:
+        eventually(lambda: smi.event_inited == {0})
+        # second gim PF + VF appears
+        h.add_gpu("0000:21:00.0", driver="gim", iommu_group="111")
+        h.add_vf("0000:21:02.0", pf_bdf="0000:21:00.0",
+                 iommu_group="121")
+        reg2 = discovery.discover(base_path=h.pci)
+        plugin.update_registry(
+            reg2, build_kubelet_devices(reg2.vf_map["75b3"]))
+        eventually(lambda: smi.event_inited == {0, 1})
+        smi.push(1, EVT_GPU_PRE_RESET, "new pf fault")
+        eventually(lambda: {d.ID: d.health
+                            for d in plugin.devices_snapshot()} == {
+            "0000:20:02.0": _dpapi.HEALTHY,
+            "0000:21:02.0": _dpapi.UNHEALTHY})
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet.stop()
