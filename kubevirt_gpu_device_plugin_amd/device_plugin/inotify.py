@@ -23,6 +23,7 @@ IN_MOVED_FROM = 0x00000040
 IN_MOVED_TO = 0x00000080
 IN_DELETE_SELF = 0x00000400
 IN_IGNORED = 0x00008000
+IN_Q_OVERFLOW = 0x00004000  # kernel dropped events (wd == -1)
 
 _EVENT_HDR = struct.Struct("iIII")  # wd, mask, cookie, len
 

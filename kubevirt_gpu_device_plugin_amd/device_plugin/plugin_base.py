@@ -294,6 +294,15 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         Overridden by subclasses that know the registry."""
         return {}
 
+    def _resync_health(self, group_to_ids):
+        """Ground-truth pass: health = does the group's vfio node
+        exist right now."""
+        for group, ids in group_to_ids.items():
+            exists = os.path.exists(
+                os.path.join(self.config.vfio_dir, group))
+            self.set_health(ids, dpapi.HEALTHY if exists
+                            else dpapi.UNHEALTHY)
+
     def _health_loop(self):
         """inotify loop: vfio node create/remove → health flips; removal
         of our own socket → kubelet restarted → full server restart
@@ -314,6 +323,15 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                 # recomputed per batch: device lists change in place on
                 # hotplug rescans (update_devices)
                 group_to_ids = self._group_to_ids()
+                if any(ev.mask & inotify.IN_Q_OVERFLOW
+                       for ev in events):
+                    # kernel dropped events (flip storm): re-derive
+                    # health from live node state instead of trusting
+                    # the partial stream
+                    log.warning("[%s] inotify queue overflow; "
+                                "resyncing health from /dev/vfio",
+                                self.device_name)
+                    self._resync_health(group_to_ids)
                 for ev in events:
                     if self._should_exit():
                         return

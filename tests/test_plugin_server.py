@@ -344,3 +344,22 @@ def test_preferred_allocation_avoids_shared_group(rig, synthetic_host):
             allocation_size=1)]))
     assert list(resp.container_responses[0].deviceIDs) == ["0000:2f:00.0"]
     ch.close()
+
+
+def test_resync_health_from_node_state(rig, synthetic_host):
+    """After an inotify overflow the health map is re-derived from the
+    actual /dev/vfio node state (exercised directly — overflow itself
+    needs a kernel-queue-full storm)."""
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    h.add_gpu("0000:2f:00.0", iommu_group="41")
+    _, plugin, _ = rig()
+    import os
+    os.remove(h.vfio_dir + "/41")
+    plugin._resync_health(plugin._group_to_ids())
+    assert {d.ID: d.health for d in plugin.devices_snapshot()} == {
+        "0000:0c:00.0": "Healthy", "0000:2f:00.0": "Unhealthy"}
+    h.add_vfio_node("41")
+    plugin._resync_health(plugin._group_to_ids())
+    assert all(d.health == "Healthy"
+               for d in plugin.devices_snapshot())
