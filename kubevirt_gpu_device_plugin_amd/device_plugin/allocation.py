@@ -33,6 +33,7 @@ def preferred_allocation(available_ids, must_include_ids, size,
     from env, generic_device_plugin.go:414-420).
     Raises ``ValueError`` when must-include exceeds ``size``.
     """
+    size = max(0, int(size))  # a nonsensical negative size means "none"
     if island_of is None:
         island_of = lambda _id: -1  # noqa: E731
     if group_size_of is not None:
