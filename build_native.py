@@ -32,11 +32,6 @@ def newer(src, dst):
             or os.path.getmtime(src) > os.path.getmtime(dst))
 
 
-def run(cmd):
-    print("+", " ".join(cmd), flush=True)
-    subprocess.check_call(cmd)
-
-
 def build(force=False):
     incs = pybind11_includes()
     jobs = []
