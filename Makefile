@@ -46,3 +46,12 @@ update-pcidb:
 
 notices:
 	$(PYTHON) tools/generate_notices.py > THIRD_PARTY_NOTICES.md
+
+burnin: build
+	$(PYTHON) tools/gpu_burnin.py --seconds 60
+
+soak: build
+	$(PYTHON) tools/daemon_soak.py --seconds 120
+
+diag: build
+	$(PYTHON) tools/dump_node_info.py
