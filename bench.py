@@ -54,6 +54,8 @@ def main():
         import torch
         import torch.distributed as tdist
         dist = tdist
+        from datetime import timedelta
+
         # RCCL only when every rank actually has its own GPU
         backend = ("nccl" if torch.cuda.is_available()
                    and torch.cuda.device_count() >= world_size
@@ -62,7 +64,9 @@ def main():
             local_rank = int(os.environ.get("LOCAL_RANK", rank))
             device = torch.device("cuda", local_rank)
             torch.cuda.set_device(device)
-        dist.init_process_group(backend=backend)
+        # bounded init: a dead rank must fail the run fast, not hang it
+        dist.init_process_group(backend=backend,
+                                timeout=timedelta(seconds=300))
 
     def barrier_sync():
         if dist is not None:
