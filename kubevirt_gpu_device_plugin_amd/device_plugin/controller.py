@@ -44,6 +44,21 @@ def resolve_name(device_id, pci_ids_path=None):
     return name
 
 
+def log_vf_partitioning(registry):
+    """Surface the HBM split per VF (VF partitioning sized
+    against 288 GB HBM3E per MI355X — 8 VFs ⇒ 36 GB each)."""
+    mi355x_vf_ids = {"75b0", "75b3"}
+    for pf, vfs in sorted(registry.pf_vf_map.items()):
+        vf_types = {d.device_id for devs in registry.vf_map.values()
+                    for d in devs if d.parent_pf == pf}
+        if vf_types & mi355x_vf_ids:
+            per_vf_gib = consts.MI355X_HBM_BYTES / len(vfs) / 1024**3
+            log.info("PF %s: %d VFs → %.1f GiB HBM3E per VF",
+                     pf, len(vfs), per_vf_gib)
+        else:
+            log.info("PF %s: %d VFs", pf, len(vfs))
+
+
 class Controller:
     """Owns discovery output and the per-resource plugin servers."""
 
@@ -74,6 +89,7 @@ class Controller:
                  {k: [d.addr for d in v]
                   for k, v in self.registry.vf_map.items()})
         log.info("pf→vf map: %s", self.registry.pf_vf_map)
+        log_vf_partitioning(self.registry)
 
         self.plugins = [
             self._make_plugin(name, kind, devs, island_of)

@@ -114,3 +114,21 @@ def test_unknown_device_id_uses_raw_hex(synthetic_host):
     plugins = ctrl.create_plugins()
     assert [p.device_name for p in plugins] == ["beef"]
     assert plugins[0].resource_name == "amd.com/beef"
+
+
+def test_vf_partition_logging(synthetic_host, caplog):
+    import logging
+    from kubevirt_gpu_device_plugin_amd.device_plugin.controller import (
+        log_vf_partitioning,
+    )
+    from kubevirt_gpu_device_plugin_amd.device_plugin import discovery
+    h = synthetic_host
+    h.add_gpu("0000:20:00.0", driver="gim", iommu_group="110")
+    for v in range(8):
+        h.add_vf("0000:20:02.%d" % v, pf_bdf="0000:20:00.0",
+                 iommu_group=str(120 + v))
+    reg = discovery.discover(base_path=h.pci)
+    with caplog.at_level(logging.INFO):
+        log_vf_partitioning(reg)
+    assert any("8 VFs" in r.message and "36.0 GiB" in r.message
+               for r in caplog.records)
