@@ -151,20 +151,23 @@ def test_smi_watcher_survives_restart_one_thread(vf_rig):
     import threading as _threading
     smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u0"}])
     h, pf, plugin, kubelet = vf_rig(n_vfs=2, smi=smi)
-    eventually(lambda: smi.event_inited == {0})
+    eventually(lambda: smi.event_inited == {0}, timeout=10.0)
     plugin.restart()
-    eventually(lambda: smi.event_inited == {0})  # re-armed
+    eventually(lambda: smi.event_inited == {0}, timeout=10.0)  # re-armed
 
     # exactly one live thread owned by THIS rig's watcher (other tests
     # may run their own shared watchers in the same process)
     watcher = plugin._smi_watcher
     eventually(lambda: watcher._thread is not None
-               and watcher._thread.is_alive(), timeout=8.0)
+               and watcher._thread.is_alive(), timeout=10.0)
     assert _threading.active_count() > 0  # sanity
 
     smi.push(0, EVT_GPU_PRE_RESET, "after restart")
+    # generous window: a restart's gRPC handshakes plus an
+    # oversubscribed runner can stretch delivery well past 5 s
     eventually(lambda: all(
-        d.health == dpapi.UNHEALTHY for d in plugin.devices_snapshot()))
+        d.health == dpapi.UNHEALTHY
+        for d in plugin.devices_snapshot()), timeout=20.0)
 
 
 def test_two_vf_plugins_share_one_watcher(synthetic_host):
