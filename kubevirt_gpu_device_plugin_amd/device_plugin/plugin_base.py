@@ -102,9 +102,15 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         # before the watcher is armed would be missed (the reference
         # registers first, generic_device_plugin.go:246-252 — benign
         # there only because fsnotify setup races the same way).
+        # The thread binds THIS generation's term/stop events at spawn:
+        # reading self._term lazily would let a thread from the previous
+        # generation survive a restart (its term was replaced before it
+        # woke) and react to our own stop()'s socket removal with a
+        # spurious second restart.
         self._watch_armed = threading.Event()
         self._health_thread = threading.Thread(
             target=self._health_loop_guard,
+            args=(self._term, self._stop),
             name="health-%s" % self.device_name, daemon=True)
         self._health_thread.start()
         self._watch_armed.wait(timeout=2.0)
@@ -129,7 +135,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             server.stop(grace=None)
             self._cleanup_socket()
 
-    def restart(self):
+    def restart(self, generation_term=None):
         """Full re-handshake after a kubelet restart
         (reference: restart, generic_device_plugin.go:275-286).
 
@@ -145,6 +151,11 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                 # the daemon is shutting down concurrently — a restart
                 # here would resurrect a server the controller just
                 # stopped
+                return
+            if generation_term is not None and generation_term.is_set():
+                # the caller belongs to an already-retired server
+                # generation (a restart completed while it was waking) —
+                # a second restart would tear down the live server
                 return
             log.info("restarting %s device plugin server",
                      self.device_name)
@@ -283,9 +294,9 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
 
     # ---- health ---------------------------------------------------------
 
-    def _health_loop_guard(self):
+    def _health_loop_guard(self, term, stop):
         try:
-            self._health_loop()
+            self._health_loop(term, stop)
         except Exception:
             log.exception("[%s] health loop failed", self.device_name)
 
@@ -303,10 +314,16 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             self.set_health(ids, dpapi.HEALTHY if exists
                             else dpapi.UNHEALTHY)
 
-    def _health_loop(self):
+    def _health_loop(self, term, stop):
         """inotify loop: vfio node create/remove → health flips; removal
         of our own socket → kubelet restarted → full server restart
-        (reference: healthCheck, generic_device_plugin.go:619-697)."""
+        (reference: healthCheck, generic_device_plugin.go:619-697).
+
+        ``term``/``stop`` are this server generation's lifetime events,
+        bound at thread spawn (see start())."""
+        def should_exit():
+            return term.is_set() or (stop is not None and stop.is_set())
+
         sock_base = os.path.basename(self.socket_path)
         with inotify.Watcher() as w:
             w.add_watch(self.config.device_plugin_dir)
@@ -316,7 +333,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                 log.warning("[%s] cannot watch %s: %s", self.device_name,
                             self.config.vfio_dir, e)
             self._watch_armed.set()
-            while not self._should_exit():
+            while not should_exit():
                 events = w.read_events(timeout_s=0.2)
                 if not events:
                     continue
@@ -333,7 +350,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                                 self.device_name)
                     self._resync_health(group_to_ids)
                 for ev in events:
-                    if self._should_exit():
+                    if should_exit():
                         return
                     path = w.path_of(ev.wd)
                     if (path == self.config.device_plugin_dir
@@ -342,7 +359,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                                            | inotify.IN_MOVED_FROM)):
                         log.info("[%s] socket removed; kubelet likely "
                                  "restarted", self.device_name)
-                        self.restart()
+                        self.restart(generation_term=term)
                         return
                     if path == self.config.vfio_dir \
                             and ev.name in group_to_ids:
