@@ -31,6 +31,9 @@ def proc_stats(pid):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=120.0)
+    ap.add_argument("--clients", type=int, default=1,
+                    help="parallel Allocate clients (kubelet admits "
+                         "pods concurrently)")
     args = ap.parse_args()
 
     with tempfile.TemporaryDirectory() as tmp:
@@ -49,40 +52,55 @@ def main():
             pid = plugin.proc.pid
             time.sleep(1.0)
             rss0, fds0 = proc_stats(pid)
-            allocs = errors = rescans = flips = 0
             deadline = time.time() + args.seconds
-            last_hup = time.time()
             rng = random.Random(42)
+            counters = {"allocs": 0, "errors": 0}
+            import threading
+            clock = threading.Lock()
+
+            def client(seed):
+                crng = random.Random(seed)
+                cch, cstub = dial_plugin(sock, timeout=10)
+                a = e = 0
+                while time.time() < deadline:
+                    try:
+                        bdf = crng.choice(devices)
+                        cstub.Allocate(dpapi.AllocateRequest(
+                            container_requests=[
+                                dpapi.ContainerAllocateRequest(
+                                    devicesIDs=[bdf])]), timeout=5)
+                        a += 1
+                    except Exception:
+                        e += 1
+                        time.sleep(0.1)
+                cch.close()
+                with clock:
+                    counters["allocs"] += a
+                    counters["errors"] += e
+
+            threads = [threading.Thread(target=client, args=(i,),
+                                        daemon=True)
+                       for i in range(args.clients)]
+            for t in threads:
+                t.start()
+
+            rescans = flips = 0
+            last_hup = time.time()
             while time.time() < deadline:
-                try:
-                    bdf = rng.choice(devices)
-                    stub.Allocate(dpapi.AllocateRequest(
-                        container_requests=[
-                            dpapi.ContainerAllocateRequest(
-                                devicesIDs=[bdf])]), timeout=5)
-                    allocs += 1
-                except Exception as e:
-                    # rescan restarts the server; transient UNAVAILABLE
-                    # is expected around a HUP
-                    errors += 1
-                    ch.close()
-                    for _ in range(50):
-                        try:
-                            ch, stub = dial_plugin(sock, timeout=5)
-                            break
-                        except Exception:
-                            time.sleep(0.2)
-                if allocs % 20 == 0:
-                    g = rng.choice([str(100 + i) for i in range(8)])
-                    host.remove_vfio_node(g)
-                    host.add_vfio_node(g)
-                    flips += 1
+                g = rng.choice([str(100 + i) for i in range(8)])
+                host.remove_vfio_node(g)
+                host.add_vfio_node(g)
+                flips += 1
                 if time.time() - last_hup > 10:
                     plugin.proc.send_signal(signal.SIGHUP)
                     rescans += 1
                     last_hup = time.time()
                     # diff-rescan: same types ⇒ same socket, no
                     # re-registration — traffic just continues
+                time.sleep(0.02)
+            for t in threads:
+                t.join(timeout=30)
+            allocs, errors = counters["allocs"], counters["errors"]
             rss1, fds1 = proc_stats(pid)
             ch.close()
             print("soak %.0fs: %d allocs, %d transient errors, "
