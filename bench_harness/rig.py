@@ -109,14 +109,14 @@ def measure_allocate(n_gpus, steps, warmup, iommufd=False,
                     by_pf.setdefault(d.rsplit(".", 1)[0][:7], []) \
                         .append(d)
                 reqs = [dpapi.AllocateRequest(container_requests=[
-                    dpapi.ContainerAllocateRequest(devicesIDs=vfs)])
+                    dpapi.ContainerAllocateRequest(devices_ids=vfs)])
                     for vfs in by_pf.values()]
             elif allocate_all:
                 reqs = [dpapi.AllocateRequest(container_requests=[
-                    dpapi.ContainerAllocateRequest(devicesIDs=devices)])]
+                    dpapi.ContainerAllocateRequest(devices_ids=devices)])]
             else:
                 reqs = [dpapi.AllocateRequest(container_requests=[
-                    dpapi.ContainerAllocateRequest(devicesIDs=[d])])
+                    dpapi.ContainerAllocateRequest(devices_ids=[d])])
                     for d in devices]
 
             for i in range(warmup):

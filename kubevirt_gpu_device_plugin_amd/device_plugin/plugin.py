@@ -124,7 +124,7 @@ class GenericDevicePlugin(DevicePluginBase):
             # generic_device_plugin.go:383-410 — that is where its
             # Allocate latency lives).
             plan = []
-            for bdf in req.devicesIDs:
+            for bdf in req.devices_ids:
                 group = self.registry.bdf_to_iommu.get(bdf)
                 members = self.registry.iommu_map.get(group, [])
                 if group is None or not members \

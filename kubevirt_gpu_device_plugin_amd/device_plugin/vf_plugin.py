@@ -115,7 +115,7 @@ class VfDevicePlugin(GenericDevicePlugin):
         """
         mine = {d.ID for d in self._devs}
         for req in request.container_requests:
-            for bdf in req.devicesIDs:
+            for bdf in req.devices_ids:
                 if bdf not in mine:
                     context.abort(
                         grpc.StatusCode.INVALID_ARGUMENT,

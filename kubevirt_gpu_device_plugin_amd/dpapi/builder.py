@@ -106,14 +106,14 @@ def build_file_descriptor_proto():
         _field("deviceIDs", 1, S, R, json_name="deviceIDs"),
     ])
     _message(fdp, "PreStartContainerRequest", [
-        _field("devicesIDs", 1, S, R, json_name="devicesIDs"),
+        _field("devices_ids", 1, S, R),
     ])
     _message(fdp, "PreStartContainerResponse", [])
     _message(fdp, "AllocateRequest", [
         _field("container_requests", 1, M, R, t("ContainerAllocateRequest")),
     ])
     _message(fdp, "ContainerAllocateRequest", [
-        _field("devicesIDs", 1, S, R, json_name="devicesIDs"),
+        _field("devices_ids", 1, S, R),
     ])
     _message(fdp, "AllocateResponse", [
         _field("container_responses", 1, M, R, t("ContainerAllocateResponse")),

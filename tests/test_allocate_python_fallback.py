@@ -47,7 +47,7 @@ def test_fallback_happy_path(fallback_rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     c = resp.container_responses[0]
     assert [d.host_path for d in c.devices] == [
         h.vfio_dir + "/vfio", h.vfio_dir + "/40"]
@@ -62,7 +62,7 @@ def test_fallback_iommufd_spec_order(fallback_rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     assert [d.host_path for d in resp.container_responses[0].devices] \
         == [h.vfio_dir + "/devices/vfio3", h.vfio_dir + "/vfio",
             h.vfio_dir + "/40", h.iommu_dev]
@@ -80,7 +80,7 @@ def test_fallback_toctou_rejected(fallback_rig, synthetic_host):
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     ch.close()
 
@@ -94,6 +94,6 @@ def test_fallback_missing_cdev_internal(fallback_rig, synthetic_host):
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INTERNAL
     ch.close()

@@ -48,7 +48,7 @@ def test_config1_single_fake_device_plumbing(synthetic_host):
         ch, stub = dial_plugin(p.socket_path)
         resp = stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
         c = resp.container_responses[0]
         assert c.envs["PCI_RESOURCE_AMD_COM_INSTINCT_MI355X"] \
             == "0000:0c:00.0"
@@ -75,7 +75,7 @@ def test_config3_eight_gpu_passthrough(synthetic_host):
         for g in range(8):
             resp = stub.Allocate(dpapi.AllocateRequest(
                 container_requests=[dpapi.ContainerAllocateRequest(
-                    devicesIDs=["0000:%02x:00.0" % (0x10 + g)])]))
+                    devices_ids=["0000:%02x:00.0" % (0x10 + g)])]))
             assert len(resp.container_responses[0].devices) == 2
         ch.close()
     finally:

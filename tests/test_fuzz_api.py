@@ -48,12 +48,12 @@ def test_allocate_weird_ids_rejected_cleanly(live):
         with pytest.raises(grpc.RpcError) as exc:
             live.Allocate(dpapi.AllocateRequest(
                 container_requests=[dpapi.ContainerAllocateRequest(
-                    devicesIDs=[wid])]))
+                    devices_ids=[wid])]))
         assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT, wid
     # server still healthy afterwards
     resp = live.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     assert resp.container_responses[0].envs
 
 
@@ -69,7 +69,7 @@ def test_allocate_empty_and_oversized_requests(live):
     # many container requests at once
     resp = live.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])] * 64))
+            devices_ids=["0000:0c:00.0"])] * 64))
     assert len(resp.container_responses) == 64
 
 

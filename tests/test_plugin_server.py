@@ -84,7 +84,7 @@ def test_allocate_happy_path_spec_order(rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     c = resp.container_responses[0]
     assert dict(c.envs) == {
         "PCI_RESOURCE_AMD_COM_INSTINCT_MI355X": "0000:0c:00.0"}
@@ -105,7 +105,7 @@ def test_allocate_iommufd_spec_order(rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     c = resp.container_responses[0]
     assert [d.host_path for d in c.devices] == [
         h.vfio_dir + "/devices/vfio3",
@@ -127,7 +127,7 @@ def test_allocate_sibling_env_suppression(rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0"])]))
+            devices_ids=["0000:0c:00.0"])]))
     c = resp.container_responses[0]
     assert dict(c.envs) == {
         "PCI_RESOURCE_AMD_COM_INSTINCT_MI355X": "0000:0c:00.0"}
@@ -144,7 +144,7 @@ def test_allocate_unknown_bdf_rejected(rig, synthetic_host):
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:ff:00.0"])]))
+                devices_ids=["0000:ff:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     assert "unknown device: 0000:ff:00.0" in exc.value.details()
     ch.close()
@@ -163,7 +163,7 @@ def test_allocate_toctou_vendor_change_rejected(rig, synthetic_host):
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     ch.close()
 
@@ -180,7 +180,7 @@ def test_allocate_toctou_iommu_change_rejected(rig, synthetic_host):
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     ch.close()
 
@@ -193,7 +193,7 @@ def test_allocate_multi_device(rig, synthetic_host):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:00.0", "0000:2f:00.0"])]))
+            devices_ids=["0000:0c:00.0", "0000:2f:00.0"])]))
     c = resp.container_responses[0]
     assert dict(c.envs) == {"PCI_RESOURCE_AMD_COM_INSTINCT_MI355X":
                             "0000:0c:00.0,0000:2f:00.0"}
@@ -211,8 +211,8 @@ def test_allocate_env_scoped_per_container(rig, synthetic_host):
     _, plugin, _ = rig()
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(container_requests=[
-        dpapi.ContainerAllocateRequest(devicesIDs=["0000:0c:00.0"]),
-        dpapi.ContainerAllocateRequest(devicesIDs=["0000:2f:00.0"]),
+        dpapi.ContainerAllocateRequest(devices_ids=["0000:0c:00.0"]),
+        dpapi.ContainerAllocateRequest(devices_ids=["0000:2f:00.0"]),
     ]))
     envs = [dict(c.envs) for c in resp.container_responses]
     assert envs[0] == {"PCI_RESOURCE_AMD_COM_INSTINCT_MI355X":
@@ -312,7 +312,7 @@ def test_allocate_iommufd_missing_cdev_internal_error(rig,
     with pytest.raises(grpc.RpcError) as exc:
         stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:0c:00.0"])]))
+                devices_ids=["0000:0c:00.0"])]))
     assert exc.value.code() == grpc.StatusCode.INTERNAL
     assert "iommufd" in exc.value.details()
     ch.close()
@@ -323,7 +323,7 @@ def test_pre_start_container(rig, synthetic_host):
     _, plugin, _ = rig()
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.PreStartContainer(dpapi.PreStartContainerRequest(
-        devicesIDs=["0000:0c:00.0"]))
+        devices_ids=["0000:0c:00.0"]))
     assert resp == dpapi.PreStartContainerResponse()
     ch.close()
 

@@ -72,7 +72,7 @@ def test_concurrent_allocates(synthetic_host):
             bdf = bdfs[i % 8]
             resp = stub.Allocate(dpapi.AllocateRequest(
                 container_requests=[dpapi.ContainerAllocateRequest(
-                    devicesIDs=[bdf])]))
+                    devices_ids=[bdf])]))
             c = resp.container_responses[0]
             assert dict(c.envs) == {
                 "PCI_RESOURCE_AMD_COM_INSTINCT_MI355X": bdf}
@@ -146,7 +146,7 @@ def test_vf_type_validation(synthetic_host):
         with pytest.raises(grpc.RpcError) as exc:
             stub.Allocate(dpapi.AllocateRequest(
                 container_requests=[dpapi.ContainerAllocateRequest(
-                    devicesIDs=["0000:10:00.0"])]))
+                    devices_ids=["0000:10:00.0"])]))
         assert exc.value.code() == grpc.StatusCode.INVALID_ARGUMENT
         assert "is not a INSTINCT_MI355X_VF" in exc.value.details()
         ch.close()
@@ -212,7 +212,7 @@ def test_vf_plugin_with_real_smi_binding_degrades(synthetic_host):
         ch, stub = dial_plugin(plugin.socket_path)
         resp = stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:20:02.0"])]))
+                devices_ids=["0000:20:02.0"])]))
         assert resp.container_responses[0].envs
         ch.close()
     finally:
@@ -296,7 +296,7 @@ def test_rescan_updates_existing_type_in_place(synthetic_host):
         # new VF is allocatable immediately
         resp = stub.Allocate(dpapi.AllocateRequest(
             container_requests=[dpapi.ContainerAllocateRequest(
-                devicesIDs=["0000:20:02.3"])]))
+                devices_ids=["0000:20:02.3"])]))
         assert resp.container_responses[0].envs
         # health watch covers the new VF too
         h.remove_vfio_node("123")

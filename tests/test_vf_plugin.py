@@ -69,7 +69,7 @@ def test_vf_register_and_allocate(vf_rig):
     ch, stub = dial_plugin(plugin.socket_path)
     resp = stub.Allocate(dpapi.AllocateRequest(
         container_requests=[dpapi.ContainerAllocateRequest(
-            devicesIDs=["0000:0c:02.1"])]))
+            devices_ids=["0000:0c:02.1"])]))
     c = resp.container_responses[0]
     # VFs are PCI vfio devices: per-group node wiring, not the
     # reference's single shared /dev/vfio spec for mdevs

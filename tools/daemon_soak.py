@@ -68,7 +68,7 @@ def main():
                         cstub.Allocate(dpapi.AllocateRequest(
                             container_requests=[
                                 dpapi.ContainerAllocateRequest(
-                                    devicesIDs=[bdf])]), timeout=5)
+                                    devices_ids=[bdf])]), timeout=5)
                         a += 1
                     except Exception:
                         e += 1
