@@ -95,6 +95,21 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         server.start()
         self._server = server
 
+        try:
+            self._start_watch_and_register()
+        except (grpc.RpcError, grpc.FutureTimeoutError, OSError) as e:
+            # A failed start must not leak a live server: without this,
+            # the socket stays bound and the health thread stays alive,
+            # and a later rescan() would create a second server on the
+            # same socket path while the zombie can independently
+            # re-register with a stale registry.
+            log.error("[%s] error starting plugin server: %s",
+                      self.device_name, e)
+            self.stop()
+            raise
+        log.info("%s device plugin server ready", self.device_name)
+
+    def _start_watch_and_register(self):
         self._wait_for_ready()
 
         # Arm the health watcher BEFORE registering: the moment kubelet
@@ -115,14 +130,7 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         self._health_thread.start()
         self._watch_armed.wait(timeout=2.0)
 
-        try:
-            self.register()
-        except grpc.RpcError as e:
-            log.error("[%s] error registering with kubelet: %s",
-                      self.device_name, e)
-            self.stop()
-            raise
-        log.info("%s device plugin server ready", self.device_name)
+        self.register()
 
     def stop(self):
         with self._lifecycle:
