@@ -139,10 +139,17 @@ class SharedSmiWatcher:
         def unsubscribe(self):
             self._owner._unsubscribe(self._handle)
 
+    def _available(self):
+        if self._smi is not None:
+            return self._smi.available()
+        return is_available()
+
     def subscribe(self, pf_bdfs, on_unhealthy, on_healthy):
         """Returns a Subscription, or None when AMD-SMI is unavailable
-        on this host (callers degrade to sysfs-only health)."""
-        if self._smi is None and not is_available():
+        on this host (callers degrade to sysfs-only health).  Checked
+        per call, not cached: a library that appears after daemon start
+        is picked up by the next (re)subscription attempt."""
+        if not self._available():
             return None
         with self._mgmt:
             with self._lock:

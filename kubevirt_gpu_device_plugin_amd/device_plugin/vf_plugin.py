@@ -56,17 +56,47 @@ class VfDevicePlugin(GenericDevicePlugin):
 
     def start(self, stop_event):
         super().start(stop_event)
-        if self._event_watcher_factory is not None \
-                and self._event_watcher_factory() is None:
+        self._subscribe_smi(self.registry)
+
+    def stop(self):
+        sub, self._smi_sub = self._smi_sub, None
+        if sub is not None:
+            sub.unsubscribe()
+        super().stop()
+
+    def update_registry(self, registry, devices, island_of=None):
+        super().update_registry(registry, devices, island_of=island_of)
+        # PF set may have changed (new gim PFs) — resubscribe.  Always
+        # attempted, not only when a prior subscription exists: if
+        # libamd_smi was unavailable at start (subscribe returned None),
+        # a rescan is exactly when it may have become available and VF
+        # health must not silently stay sysfs-only for the daemon's
+        # lifetime.
+        self._subscribe_smi(registry)
+
+    def _smi_disabled(self):
+        """event_watcher_factory=lambda: None turns SMI health off
+        entirely (test/bench seam)."""
+        return (self._event_watcher_factory is not None
+                and self._event_watcher_factory() is None)
+
+    def _subscribe_smi(self, registry):
+        """(Re)subscribe the shared watcher for the current PF set;
+        degrades to sysfs-only health when AMD-SMI is unavailable
+        (reference contract: generic_vgpu_device_plugin.go:290-297)."""
+        if self._smi_disabled():
             log.info("[%s] SMI health disabled by configuration",
                      self.device_name)
             return
+        sub, self._smi_sub = self._smi_sub, None
+        if sub is not None:
+            sub.unsubscribe()
         watcher = self._smi_watcher
         if watcher is None:
             from ..amdsmi import events as smi_events
             watcher = smi_events.shared_watcher()
         pf_bdfs = sorted({d.parent_pf
-                          for devs in self.registry.vf_map.values()
+                          for devs in registry.vf_map.values()
                           for d in devs if d.parent_pf})
         self._smi_sub = watcher.subscribe(
             pf_bdfs,
@@ -77,32 +107,6 @@ class VfDevicePlugin(GenericDevicePlugin):
         if self._smi_sub is None:
             log.warning("[%s] AMD-SMI unavailable; VF health relies on "
                         "vfio node watching only", self.device_name)
-
-    def stop(self):
-        sub, self._smi_sub = self._smi_sub, None
-        if sub is not None:
-            sub.unsubscribe()
-        super().stop()
-
-    def update_registry(self, registry, devices, island_of=None):
-        super().update_registry(registry, devices, island_of=island_of)
-        # PF set may have changed (new gim PFs) — resubscribe
-        sub, self._smi_sub = self._smi_sub, None
-        if sub is not None:
-            sub.unsubscribe()
-            pf_bdfs = sorted({d.parent_pf
-                              for devs in registry.vf_map.values()
-                              for d in devs if d.parent_pf})
-            watcher = self._smi_watcher
-            if watcher is None:
-                from ..amdsmi import events as smi_events
-                watcher = smi_events.shared_watcher()
-            self._smi_sub = watcher.subscribe(
-                pf_bdfs,
-                on_unhealthy=lambda pf: self.set_health(
-                    self._vfs_of_pf(pf), dpapi.UNHEALTHY),
-                on_healthy=lambda pf: self.set_health(
-                    self._vfs_of_pf(pf), dpapi.HEALTHY))
 
     def Allocate(self, request, context):  # noqa: N802
         """VF allocation = passthrough allocation, plus resource-type

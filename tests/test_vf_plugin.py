@@ -206,3 +206,29 @@ def test_two_vf_plugins_share_one_watcher(synthetic_host):
         for p in plugins:
             p.stop()
         kubelet.stop()
+
+
+def test_smi_becomes_available_after_rescan(vf_rig):
+    """libamd_smi absent at start (subscribe returns None) must not be
+    permanent: a rescan retries the subscription, and once SMI is
+    available PF fault events flow to VF health (regression for the
+    lost-resubscription gap; reference degradation contract:
+    generic_vgpu_device_plugin.go:290-297)."""
+    smi = FakeSmi([{"index": 0, "bdf": "0000:0c:00.0", "uuid": "u0"}])
+    avail = [False]
+    smi.available = lambda: avail[0]
+    h, pf, plugin, kubelet = vf_rig(smi=smi)
+    assert plugin._smi_sub is None  # degraded to sysfs-only health
+
+    avail[0] = True
+    reg = discovery.discover(base_path=h.pci)
+    plugin.update_registry(reg,
+                           build_kubelet_devices(reg.vf_map["75b3"]))
+    assert plugin._smi_sub is not None
+
+    smi.push(0, EVT_GPU_PRE_RESET)
+    eventually(lambda: all(d.health == dpapi.UNHEALTHY
+                           for d in plugin.devices_snapshot()))
+    smi.push(0, EVT_GPU_POST_RESET)
+    eventually(lambda: all(d.health == dpapi.HEALTHY
+                           for d in plugin.devices_snapshot()))
