@@ -48,6 +48,13 @@ SUPPORTED_VFIO_DRIVERS = frozenset({"vfio-pci"})
 # generic_device_plugin.go:53).
 CONNECTION_TIMEOUT_S = 5.0
 
+# Period of the ground-truth health resync in the health loop: the only
+# health source while the /dev/vfio inotify watch cannot be established,
+# and a safety net against missed events otherwise (the reference goes
+# fully blind when its fsnotify setup fails,
+# generic_device_plugin.go:626-637).
+HEALTH_RESYNC_S = 30.0
+
 # Device node permissions requested from kubelet.
 DEVICE_PERMISSIONS = "mrw"
 

@@ -14,6 +14,7 @@ equivalent here.
 """
 
 import logging
+import os
 
 import grpc
 
@@ -78,6 +79,26 @@ class GenericDevicePlugin(DevicePluginBase):
                 continue
             mapping.setdefault(group, []).append(dev.ID)
         return mapping
+
+    def _resync_health(self, group_to_ids):
+        """Ground-truth health pass, cdev-aware: a device is healthy
+        when its /dev/vfio/<group> node exists, or — on hosts running
+        vfio in pure iommufd cdev mode, where group nodes are never
+        created — when its per-device /dev/vfio/devices/vfioN node
+        exists (vfioN resolved from the sysfs vfio-dev entry)."""
+        for group, ids in group_to_ids.items():
+            if os.path.exists(self._vfio_prefix + group):
+                self.set_health(ids, dpapi.HEALTHY)
+                continue
+            for bdf in ids:
+                try:
+                    vfiodev = sysfs.read_vfio_dev(
+                        self.config.pci_base, bdf)
+                    ok = os.path.exists(self._cdev_prefix + vfiodev)
+                except OSError:
+                    ok = False  # no vfio-dev entry → not cdev-bound
+                self.set_health([bdf], dpapi.HEALTHY if ok
+                                else dpapi.UNHEALTHY)
 
     # ---- Allocate -------------------------------------------------------
 

@@ -43,6 +43,10 @@ class PluginConfig:
     pci_base: str = consts.PCI_DEVICES_PATH
     namespace: str = consts.DEVICE_NAMESPACE
     connect_timeout_s: float = consts.CONNECTION_TIMEOUT_S
+    # period of the ground-truth health resync pass; it is the only
+    # health source while the /dev/vfio watch cannot be established and
+    # a belt-and-braces check against missed inotify events otherwise
+    health_resync_s: float = consts.HEALTH_RESYNC_S
 
 
 class DevicePluginBase(dpapi.DevicePluginServicer):
@@ -335,14 +339,36 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         sock_base = os.path.basename(self.socket_path)
         with inotify.Watcher() as w:
             w.add_watch(self.config.device_plugin_dir)
+            vfio_watched = False
             try:
                 w.add_watch(self.config.vfio_dir)
+                vfio_watched = True
             except OSError as e:
-                log.warning("[%s] cannot watch %s: %s", self.device_name,
-                            self.config.vfio_dir, e)
+                log.warning("[%s] cannot watch %s (%s); falling back to "
+                            "periodic ground-truth health resync",
+                            self.device_name, self.config.vfio_dir, e)
             self._watch_armed.set()
+            last_resync = time.monotonic()
             while not should_exit():
                 events = w.read_events(timeout_s=0.2)
+                # Ground-truth pass: the ONLY health source while the
+                # vfio dir is unwatchable (otherwise devices would stay
+                # Healthy forever with zero checks), and a low-frequency
+                # safety net against missed inotify events when watched.
+                if time.monotonic() - last_resync \
+                        >= self.config.health_resync_s:
+                    last_resync = time.monotonic()
+                    if not vfio_watched:
+                        # the dir may exist by now (vfio module loaded
+                        # after daemon start) — prefer event-driven
+                        try:
+                            w.add_watch(self.config.vfio_dir)
+                            vfio_watched = True
+                            log.info("[%s] /dev/vfio watch established",
+                                     self.device_name)
+                        except OSError:
+                            pass
+                    self._resync_health(self._group_to_ids())
                 if not events:
                     continue
                 # recomputed per batch: device lists change in place on
