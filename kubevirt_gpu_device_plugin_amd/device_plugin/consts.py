@@ -55,6 +55,13 @@ CONNECTION_TIMEOUT_S = 5.0
 # generic_device_plugin.go:626-637).
 HEALTH_RESYNC_S = 30.0
 
+# Kubelet-restart re-registration backoff: doubled per failed attempt
+# up to the cap, retried forever while the daemon lives (the reference
+# abandons the resource after a single failed re-register,
+# generic_device_plugin.go:688-692).
+RESTART_BACKOFF_INITIAL_S = 0.5
+RESTART_BACKOFF_MAX_S = 30.0
+
 # Device node permissions requested from kubelet.
 DEVICE_PERMISSIONS = "mrw"
 

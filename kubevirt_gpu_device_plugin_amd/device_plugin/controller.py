@@ -200,7 +200,12 @@ def initiate_device_plugin(stop_event=None, rescan_event=None,
     controller.create_plugins()
     controller.start(stop_event)
     while not stop_event.is_set():
-        stop_event.wait(0.5)
+        # woken immediately by a rescan request; the 0.5 s timeout only
+        # bounds shutdown latency (stop_event has no waiter-wakeup hook
+        # we own — callers hand us an arbitrary Event)
+        rescan_event.wait(0.5)
+        if stop_event.is_set():
+            break
         if rescan_event.is_set():
             # debounce: a gim VF burst emits one uevent per function
             stop_event.wait(0.5)

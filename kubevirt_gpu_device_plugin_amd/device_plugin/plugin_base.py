@@ -47,6 +47,10 @@ class PluginConfig:
     # health source while the /dev/vfio watch cannot be established and
     # a belt-and-braces check against missed inotify events otherwise
     health_resync_s: float = consts.HEALTH_RESYNC_S
+    # kubelet-restart re-registration backoff (doubles up to the cap;
+    # retries forever while the daemon lives)
+    restart_backoff_initial_s: float = consts.RESTART_BACKOFF_INITIAL_S
+    restart_backoff_max_s: float = consts.RESTART_BACKOFF_MAX_S
 
 
 class DevicePluginBase(dpapi.DevicePluginServicer):
@@ -173,9 +177,17 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                      self.device_name)
             stop_event = self._stop
             self.stop()
-            for attempt in range(30):
+            # Retry forever with capped exponential backoff: a resource
+            # is never abandoned while the daemon lives — a kubelet
+            # outage of any length ends with re-registration.  (The old
+            # behavior gave up after ~60 s; the reference gives up after
+            # ONE attempt, generic_device_plugin.go:688-692.)
+            backoff = self.config.restart_backoff_initial_s
+            attempt = 0
+            while True:
                 if stop_event is not None and stop_event.is_set():
                     return
+                attempt += 1
                 try:
                     self.start(stop_event)
                     return
@@ -183,15 +195,15 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                         OSError) as e:
                     log.warning(
                         "[%s] restart attempt %d failed (%s); kubelet "
-                        "may still be coming up — retrying",
-                        self.device_name, attempt + 1, e)
+                        "may still be coming up — retrying in %.1fs",
+                        self.device_name, attempt, e, backoff)
                     self.stop()
                     if stop_event is not None:
-                        stop_event.wait(2.0)  # wakes early on shutdown
+                        stop_event.wait(backoff)  # wakes on shutdown
                     else:
-                        time.sleep(2.0)
-            log.error("[%s] giving up on restart after 30 attempts",
-                      self.device_name)
+                        time.sleep(backoff)
+                    backoff = min(backoff * 2,
+                                  self.config.restart_backoff_max_s)
 
     def _cleanup_socket(self):
         try:

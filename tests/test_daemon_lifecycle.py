@@ -84,3 +84,48 @@ def test_listandwatch_reconnect(synthetic_host):
         stop.set()
         plugin.stop()
         kubelet.stop()
+
+
+def test_restart_survives_long_kubelet_outage(synthetic_host):
+    """Chaos: kubelet stays down for far more re-register attempts than
+    the old 30-attempt cap, then comes back — the plugin must still be
+    retrying (capped backoff, never abandoned) and register.
+    (Reference behavior to beat: ONE attempt then dead,
+    generic_device_plugin.go:688-692.)"""
+    import time
+
+    from tests.fixtures import eventually
+
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    cfg = h.config()
+    cfg.connect_timeout_s = 0.05
+    cfg.restart_backoff_initial_s = 0.01
+    cfg.restart_backoff_max_s = 0.02
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    _, devs = next(iter(reg.device_map.items()))
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X", build_kubelet_devices(devs), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)
+    assert kubelet.wait_register()
+    kubelet.stop()  # kubelet goes away
+    os.remove(cfg.kubelet_socket) if os.path.exists(
+        cfg.kubelet_socket) else None
+
+    t = threading.Thread(target=plugin.restart, daemon=True)
+    t.start()
+    # outage spans well over 30 failed attempts (~0.07s per attempt)
+    time.sleep(3.0)
+    assert t.is_alive(), "restart loop gave up during the outage"
+
+    kubelet2 = StubKubelet(cfg.kubelet_socket)
+    try:
+        assert kubelet2.wait_register(timeout=10.0)
+        eventually(lambda: not t.is_alive())
+        assert plugin._server is not None
+    finally:
+        stop.set()
+        plugin.stop()
+        kubelet2.stop()
