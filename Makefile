@@ -40,8 +40,10 @@ image:
 
 # Refresh the full public PCI ID database for the container image
 # (reference: Makefile:104-105).  The curated in-package table stays as
-# the built-in fallback.
+# the built-in fallback; `make image` works with or without this having
+# run (the Dockerfile falls back to the curated table).
 update-pcidb:
+	mkdir -p utils
 	wget -O utils/pci.ids.full $(PCI_IDS_URL)
 
 notices:

@@ -40,8 +40,13 @@ def _lookup_in_file(path, device_id, vendor_id):
         for line in f:
             line = line.rstrip("\n")
             if not in_vendor:
+                # vendor line = the 4-hex id at column 0 followed by
+                # whitespace (a bare startswith would also match a
+                # hypothetical longer id sharing the prefix)
                 if line.startswith(vendor_id):
-                    in_vendor = True
+                    rest = line[len(vendor_id):]
+                    if not rest or rest[0].isspace():
+                        in_vendor = True
                 continue
             if line.startswith("#"):
                 continue

@@ -125,7 +125,8 @@ class GenericDevicePlugin(DevicePluginBase):
         """
         base = self.config.pci_base
         # /dev/iommu present ⇒ iommufd cdev flow
-        iommufd = sysfs.supports_iommufd(self.config.iommu_dev)
+        iommufd = sysfs.supports_iommufd(self.config.iommu_dev,
+                                         vfio_dir=self.config.vfio_dir)
 
         response = dpapi.AllocateResponse()
         for req in request.container_requests:

@@ -75,7 +75,14 @@ def read_vfio_dev(base_path, device_address):
                             % device_address)
 
 
-def supports_iommufd(iommu_dev_path="/dev/iommu"):
+def supports_iommufd(iommu_dev_path="/dev/iommu", vfio_dir=None):
     """Host supports iommufd when /dev/iommu exists
-    (reference: supportsIOMMUFD, generic_device_plugin.go:700-709)."""
-    return os.path.exists(iommu_dev_path)
+    (reference: supportsIOMMUFD, generic_device_plugin.go:700-709) —
+    or, containerized with only /dev/vfio mounted, when the cdev
+    directory /dev/vfio/devices exists (CONFIG_VFIO_DEVICE_CDEV depends
+    on IOMMUFD, so its presence implies /dev/iommu on the host even
+    when that node is not mounted into this pod)."""
+    if os.path.exists(iommu_dev_path):
+        return True
+    return vfio_dir is not None and os.path.isdir(
+        os.path.join(vfio_dir, "devices"))

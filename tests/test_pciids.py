@@ -108,3 +108,82 @@ def test_missing_system_db_skipped(tmp_path):
         "75a3", search_paths=[str(tmp_path / "nope"),
                               pciids.BUILTIN_IDS_PATH])
     assert got == "INSTINCT_MI355X"
+
+
+# ---- search-order contract + full-public-db hardening ------------------
+
+import os as _os
+
+import pytest as _pytest
+
+REF_FULL_IDS = "/root/reference/utils/pci.ids"
+
+
+def test_search_order_full_db_miss_falls_to_curated(tmp_path):
+    """/usr/pci.ids exists but lacks 75a3 (true of public v2025.07.11):
+    the curated built-in table must still resolve it."""
+    full = tmp_path / "pci.ids"
+    full.write_text("1002  Advanced Micro Devices, Inc. [AMD/ATI]\n"
+                    "\t74a1  Aqua Vanjaram [Instinct MI300X]\n")
+    name = pciids.get_device_name(
+        "75a3", search_paths=[str(full), pciids.BUILTIN_IDS_PATH])
+    assert name == "INSTINCT_MI355X"
+
+
+def test_search_order_full_db_name_overrides_curated(tmp_path):
+    """The converse: when the full db DOES carry the id, its name wins
+    over the curated table (first hit in search order)."""
+    full = tmp_path / "pci.ids"
+    full.write_text("1002  Advanced Micro Devices, Inc. [AMD/ATI]\n"
+                    "\t75a3  Instinct MI355X rev B\n")
+    name = pciids.get_device_name(
+        "75a3", search_paths=[str(full), pciids.BUILTIN_IDS_PATH])
+    assert name == "INSTINCT_MI355X_REV_B"
+
+
+def test_vendor_line_requires_word_boundary(tmp_path):
+    """A longer id sharing the vendor prefix must not open the block."""
+    db = tmp_path / "pci.ids"
+    db.write_text("1002b  Not A Real Vendor\n"
+                  "\t75a3  Imposter Device\n"
+                  "1002  Advanced Micro Devices, Inc. [AMD/ATI]\n"
+                  "\t75a3  Instinct MI355X\n")
+    assert pciids.get_device_name(
+        "75a3", pci_ids_path=str(db)) == "INSTINCT_MI355X"
+
+
+@_pytest.mark.skipif(not _os.path.exists(REF_FULL_IDS),
+                     reason="full public pci.ids not on this host")
+class TestAgainstFullPublicDb:
+    """Hardening against the real 40k-line public database
+    (v2025.07.11, vendored by the reference)."""
+
+    def test_amd_id_resolves(self):
+        assert pciids.get_device_name(
+            "74a1", pci_ids_path=REF_FULL_IDS) == \
+            "AQUA_VANJARAM_INSTINCT_MI300X"
+
+    def test_cross_vendor_id_never_matches(self):
+        # 2331 exists under 10de (H100) and 8086-adjacent blocks but
+        # not under 1002: vendor-block scoping must return ""
+        assert pciids.get_device_name(
+            "2331", pci_ids_path=REF_FULL_IDS) == ""
+        assert pciids.get_device_name(
+            "2331", vendor_id="10de", pci_ids_path=REF_FULL_IDS) == \
+            "GH100_H100_PCIE"
+
+    def test_prefix_id_collision_guard(self):
+        # "74a" is a strict prefix of 74a0/74a1/…: the guard must not
+        # return MI300A's name for it
+        assert pciids.get_device_name(
+            "74a", pci_ids_path=REF_FULL_IDS) == ""
+
+    def test_mi355x_absent_from_public_db_needs_curated(self):
+        # the documented "hard part" (SURVEY §7.2): public v2025.07.11
+        # has no 75a3 — proves the curated fallback is load-bearing
+        assert pciids.get_device_name(
+            "75a3", pci_ids_path=REF_FULL_IDS) == ""
+        assert pciids.get_device_name(
+            "75a3", search_paths=[REF_FULL_IDS,
+                                  pciids.BUILTIN_IDS_PATH]) == \
+            "INSTINCT_MI355X"

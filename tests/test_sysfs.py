@@ -74,3 +74,21 @@ def test_read_sriov_numvfs(synthetic_host):
     with open(os.path.join(d, "sriov_numvfs"), "w") as f:
         f.write("8\n")
     assert sysfs.read_sriov_numvfs(h.pci, "0000:0c:00.0") == 8
+
+
+def test_supports_iommufd_via_cdev_dir(tmp_path):
+    """Containerized detection: /dev/iommu not mounted, but the
+    /dev/vfio/devices cdev dir (under the mounted /dev/vfio) implies
+    iommufd support on the host."""
+    from kubevirt_gpu_device_plugin_amd.device_plugin import sysfs
+
+    iommu = str(tmp_path / "iommu")
+    vfio = tmp_path / "vfio"
+    vfio.mkdir()
+    assert not sysfs.supports_iommufd(iommu, vfio_dir=str(vfio))
+    (vfio / "devices").mkdir()
+    assert sysfs.supports_iommufd(iommu, vfio_dir=str(vfio))
+    # direct /dev/iommu presence still wins on its own
+    with open(iommu, "w"):
+        pass
+    assert sysfs.supports_iommufd(iommu, vfio_dir=None)
