@@ -1,6 +1,6 @@
 """Discovery walk tests (reference: device_plugin_test.go:279-323)."""
 
-from kubevirt_gpu_device_plugin_amd.device_plugin import discovery
+from kubevirt_gpu_device_plugin_amd.device_plugin import consts, discovery
 
 
 def test_basic_discovery(synthetic_host):
@@ -135,3 +135,30 @@ def test_garbage_numa_defaults_to_zero(synthetic_host):
         f.write("not-a-number\n")
     reg = discovery.discover(base_path=h.pci, use_native=False)
     assert reg.device_map["75a3"][0].numa_node == 0
+
+
+def test_second_vfio_driver_single_extension_point(synthetic_host):
+    """A future vfio-pci-variant driver (the AMD analogue of the
+    reference's second accepted driver, nvgrace_gpu_vfio_pci,
+    device_plugin.go:75-78) is enabled by adding ONE name to
+    consts.SUPPORTED_VFIO_DRIVERS: the same set gates both the Python
+    walk and the native-scan records."""
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")                # vfio-pci
+    h.add_gpu("0000:2f:00.0", iommu_group="41",
+              driver="vfio-pci-amdfuture")                     # variant
+
+    for use_native in (False, None):
+        reg = discovery.discover(base_path=h.pci,
+                                 use_native=use_native)
+        assert [d.addr for d in reg.device_map["75a3"]] == \
+            ["0000:0c:00.0"], "unknown driver must be skipped"
+
+        extended = consts.SUPPORTED_VFIO_DRIVERS | \
+            {"vfio-pci-amdfuture"}
+        reg = discovery.discover(base_path=h.pci,
+                                 supported_drivers=extended,
+                                 use_native=use_native)
+        assert [d.addr for d in reg.device_map["75a3"]] == \
+            ["0000:0c:00.0", "0000:2f:00.0"], \
+            "one-place extension must cover this scan path"
