@@ -17,8 +17,11 @@ test: build
 test-gpu: build
 	$(PYTHON) -m pytest tests/ -q -m gpu
 
+# full BASELINE scaling curve: 1/2/4/8 GPUs (+ the iommufd/VF config)
 bench: build
 	$(PYTHON) bench.py --gpus 1 --steps 200 --warmup 20
+	$(PYTHON) bench.py --gpus 2 --steps 200 --warmup 20
+	$(PYTHON) bench.py --gpus 4 --steps 200 --warmup 20
 	$(PYTHON) bench.py --gpus 8 --steps 200 --warmup 20 --iommufd --vf-check
 
 coverage: build
