@@ -34,6 +34,13 @@ def main():
     ap.add_argument("--clients", type=int, default=1,
                     help="parallel Allocate clients (kubelet admits "
                          "pods concurrently)")
+    ap.add_argument("--kubelet-restarts", action="store_true",
+                    help="chaos: periodically kill the stub kubelet, "
+                         "wipe the plugin socket (what a restarting "
+                         "kubelet does) and bring it back after a "
+                         "random outage — exercises the socket-removal "
+                         "restart + capped-backoff re-registration "
+                         "path under live Allocate traffic")
     args = ap.parse_args()
 
     with tempfile.TemporaryDirectory() as tmp:
@@ -84,8 +91,8 @@ def main():
             for t in threads:
                 t.start()
 
-            rescans = flips = 0
-            last_hup = time.time()
+            rescans = flips = kubelet_restarts = 0
+            last_hup = last_kr = time.time()
             while time.time() < deadline:
                 g = rng.choice([str(100 + i) for i in range(8)])
                 host.remove_vfio_node(g)
@@ -97,6 +104,20 @@ def main():
                     last_hup = time.time()
                     # diff-rescan: same types ⇒ same socket, no
                     # re-registration — traffic just continues
+                if args.kubelet_restarts \
+                        and time.time() - last_kr > 15:
+                    kubelet.stop()
+                    try:  # a restarting kubelet wipes plugin sockets
+                        os.remove(sock)
+                    except FileNotFoundError:
+                        pass
+                    time.sleep(rng.uniform(0.5, 3.0))  # outage window
+                    kubelet = StubKubelet(cfg.kubelet_socket)
+                    assert kubelet.wait_register(30), \
+                        "plugin did not re-register after kubelet " \
+                        "restart"
+                    kubelet_restarts += 1
+                    last_kr = time.time()
                 time.sleep(0.02)
             for t in threads:
                 t.join(timeout=30)
@@ -104,8 +125,9 @@ def main():
             rss1, fds1 = proc_stats(pid)
             ch.close()
             print("soak %.0fs: %d allocs, %d transient errors, "
-                  "%d rescans, %d health flips" %
-                  (args.seconds, allocs, errors, rescans, flips))
+                  "%d rescans, %d health flips, %d kubelet restarts" %
+                  (args.seconds, allocs, errors, rescans, flips,
+                   kubelet_restarts))
             print("daemon RSS %d→%d kB (Δ%+d), fds %d→%d" %
                   (rss0, rss1, rss1 - rss0, fds0, fds1))
             grew = rss1 - rss0 > 20_000 or fds1 - fds0 > 10
