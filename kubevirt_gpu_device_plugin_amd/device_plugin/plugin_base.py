@@ -73,6 +73,12 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         # serializes start/stop/restart between the controller thread
         # and the health thread's kubelet-restart path
         self._lifecycle = threading.RLock()
+        # count of threads currently blocked in stop() waiting for
+        # _lifecycle: the unbounded restart retry loop aborts when one
+        # appears, so a rescan retiring this resource (or any external
+        # stop) is never blocked behind a kubelet outage
+        self._stop_waiters = 0
+        self._stop_waiters_lock = threading.Lock()
 
     # ---- lifecycle ------------------------------------------------------
 
@@ -141,15 +147,21 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
         self.register()
 
     def stop(self):
-        with self._lifecycle:
-            if self._server is None:
-                return
-            self._term.set()
-            with self._lock:
-                self._lock.notify_all()
-            server, self._server = self._server, None
-            server.stop(grace=None)
-            self._cleanup_socket()
+        with self._stop_waiters_lock:
+            self._stop_waiters += 1
+        try:
+            with self._lifecycle:
+                if self._server is None:
+                    return
+                self._term.set()
+                with self._lock:
+                    self._lock.notify_all()
+                server, self._server = self._server, None
+                server.stop(grace=None)
+                self._cleanup_socket()
+        finally:
+            with self._stop_waiters_lock:
+                self._stop_waiters -= 1
 
     def restart(self, generation_term=None):
         """Full re-handshake after a kubelet restart
@@ -184,8 +196,16 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
             # ONE attempt, generic_device_plugin.go:688-692.)
             backoff = self.config.restart_backoff_initial_s
             attempt = 0
+
+            def abort():
+                # daemon shutting down, or another thread is blocked in
+                # stop() for this plugin (rescan retiring the resource)
+                # — retrying further would hold _lifecycle against it
+                return ((stop_event is not None and stop_event.is_set())
+                        or self._stop_waiters > 0)
+
             while True:
-                if stop_event is not None and stop_event.is_set():
+                if abort():
                     return
                 attempt += 1
                 try:
@@ -198,10 +218,14 @@ class DevicePluginBase(dpapi.DevicePluginServicer):
                         "may still be coming up — retrying in %.1fs",
                         self.device_name, attempt, e, backoff)
                     self.stop()
-                    if stop_event is not None:
-                        stop_event.wait(backoff)  # wakes on shutdown
-                    else:
-                        time.sleep(backoff)
+                    deadline = time.monotonic() + backoff
+                    while time.monotonic() < deadline:
+                        if abort():
+                            return
+                        if stop_event is not None:
+                            stop_event.wait(0.1)
+                        else:
+                            time.sleep(0.1)
                     backoff = min(backoff * 2,
                                   self.config.restart_backoff_max_s)
 

@@ -129,3 +129,43 @@ def test_restart_survives_long_kubelet_outage(synthetic_host):
         stop.set()
         plugin.stop()
         kubelet2.stop()
+
+
+def test_concurrent_stop_aborts_restart_retry_loop(synthetic_host):
+    """A rescan retiring this resource calls stop() while the restart
+    loop is mid-outage: stop() must not be blocked behind the unbounded
+    retry — the loop aborts promptly and the plugin stays stopped."""
+    import time
+
+    from tests.fixtures import eventually
+
+    h = synthetic_host
+    h.add_gpu("0000:0c:00.0", iommu_group="40")
+    cfg = h.config()
+    cfg.connect_timeout_s = 0.2
+    cfg.restart_backoff_initial_s = 0.2
+    cfg.restart_backoff_max_s = 0.4
+    kubelet = StubKubelet(cfg.kubelet_socket)
+    reg = discovery.discover(base_path=h.pci)
+    _, devs = next(iter(reg.device_map.items()))
+    plugin = GenericDevicePlugin(
+        "INSTINCT_MI355X", build_kubelet_devices(devs), reg, config=cfg)
+    stop = threading.Event()
+    plugin.start(stop)
+    assert kubelet.wait_register()
+    kubelet.stop()
+
+    t = threading.Thread(target=plugin.restart, daemon=True)
+    t.start()
+    time.sleep(0.5)  # let the retry loop begin failing attempts
+    assert t.is_alive()
+
+    t0 = time.monotonic()
+    plugin.stop()
+    stopped_in = time.monotonic() - t0
+    assert stopped_in < 2.0, (
+        "stop() blocked %.1fs behind the restart retry loop"
+        % stopped_in)
+    eventually(lambda: not t.is_alive())
+    assert plugin._server is None
+    assert not os.path.exists(plugin.socket_path)
