@@ -114,3 +114,33 @@ def test_decoder_roundtrip_sanity():
     topo = wc.decode(by[3][1])
     node = wc.decode(wc.fields(topo, 1)[0])
     assert node == [(1, VARINT, 300)]  # multi-byte varint exercised
+
+
+def test_decoder_fixed_width_and_errors():
+    """Decoder edge cases: I64/I32 wire types, truncated LEN, overlong
+    varint — none appear in api.proto traffic, but the decoder must
+    fail loudly rather than misparse if they ever do."""
+    import struct
+
+    import pytest
+
+    # field 1, wire type 1 (I64) = tag 0x09 + 8 bytes
+    items = wc.decode(b"\x09" + struct.pack("<d", 2.5))
+    assert items == [(1, wc.WIRETYPE_I64, struct.pack("<d", 2.5))]
+    # field 2, wire type 5 (I32) = tag 0x15 + 4 bytes
+    items = wc.decode(b"\x15" + struct.pack("<f", 1.0))
+    assert items == [(2, wc.WIRETYPE_I32, struct.pack("<f", 1.0))]
+    # truncated LEN payload
+    with pytest.raises(ValueError):
+        wc.decode(b"\x0a\x05abc")
+    # unsupported wire type (3 = deprecated group start)
+    with pytest.raises(ValueError):
+        wc.decode(b"\x0b")
+    # overlong varint (>10 bytes of continuation)
+    with pytest.raises(ValueError):
+        wc.decode(b"\x08" + b"\x80" * 11)
+    # truncated fixed-width payloads
+    with pytest.raises(ValueError):
+        wc.decode(b"\x09\x00\x00")
+    with pytest.raises(ValueError):
+        wc.decode(b"\x15\x00")

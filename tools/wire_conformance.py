@@ -54,8 +54,12 @@ def decode(data):
             pos += ln
         elif wt == WIRETYPE_I64:
             val, pos = data[pos:pos + 8], pos + 8
+            if len(val) != 8:
+                raise ValueError("truncated I64 field %d" % field)
         elif wt == WIRETYPE_I32:
             val, pos = data[pos:pos + 4], pos + 4
+            if len(val) != 4:
+                raise ValueError("truncated I32 field %d" % field)
         else:
             raise ValueError("unsupported wire type %d" % wt)
         out.append((field, wt, val))
