@@ -144,3 +144,46 @@ def test_decoder_fixed_width_and_errors():
         wc.decode(b"\x09\x00\x00")
     with pytest.raises(ValueError):
         wc.decode(b"\x15\x00")
+
+
+def test_decoder_agrees_with_runtime_on_random_messages():
+    """Property check: for randomized AllocateResponse trees, the
+    independent decoder recovers exactly the fields the protobuf
+    runtime serialized (numbers, multiplicity, string payloads)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from kubevirt_gpu_device_plugin_amd import dpapi
+
+    text = st.text(
+        alphabet=st.characters(min_codepoint=32, max_codepoint=126),
+        max_size=20)
+
+    specs = st.lists(
+        st.tuples(text, text, st.sampled_from(["r", "rw", "mrw"])),
+        max_size=4)
+
+    @settings(max_examples=200, deadline=None)
+    @given(envs=st.dictionaries(text.filter(bool), text, max_size=4),
+           specs=specs)
+    def check(envs, specs):
+        car = dpapi.ContainerAllocateResponse(
+            envs=envs,
+            devices=[dpapi.DeviceSpec(container_path=c, host_path=h,
+                                      permissions=p)
+                     for c, h, p in specs])
+        items = wc.decode(car.SerializeToString())
+        got_envs = {}
+        for e in wc.fields(items, 1):
+            kv = {f: v for f, _, v in wc.decode(e)}
+            got_envs[kv.get(1, b"").decode()] = kv.get(2, b"").decode()
+        assert got_envs == dict(envs)
+        got_specs = []
+        for s in wc.fields(items, 3):
+            kv = {f: v for f, _, v in wc.decode(s)}
+            got_specs.append((kv.get(1, b"").decode(),
+                              kv.get(2, b"").decode(),
+                              kv.get(3, b"").decode()))
+        assert got_specs == [tuple(x) for x in specs]
+
+    check()
